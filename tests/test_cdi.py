@@ -1,0 +1,115 @@
+"""CDI layer tests, incl. the Kata contract golden file.
+
+Reference behavior being matched: generateCDISpec
+(pkg/device_plugin/device_plugin.go:55-80) + cdi/spec.go writer.
+"""
+import os
+
+import pytest
+import yaml
+
+from kata_xpu_device_plugin_amd.cdi import (
+    build_spec,
+    parse_qualified_name,
+    qualified_name,
+    write_spec,
+)
+from kata_xpu_device_plugin_amd.cdi.spec import read_spec, spec_path
+from kata_xpu_device_plugin_amd.discovery import scan_node
+from kata_xpu_device_plugin_amd.testing.mocknode import make_mock_node
+
+
+def test_qualified_name_roundtrip():
+    qn = qualified_name("amd.com/gpu", "70")
+    assert qn == "amd.com/gpu=70"
+    assert parse_qualified_name(qn) == ("amd.com/gpu", "70")
+    with pytest.raises(ValueError):
+        parse_qualified_name("no-equals")
+    with pytest.raises(ValueError):
+        parse_qualified_name("badkind=70")
+
+
+def test_build_spec_shape(mock_cfg):
+    inv = scan_node(mock_cfg)
+    spec = build_spec(inv, "amd.com/gpu", mock_cfg.dev_root)
+    assert spec.cdi_version == "0.8.0"
+    assert spec.kind == "amd.com/gpu"
+    assert len(spec.devices) == 8
+    d0 = spec.devices[0]
+    assert d0.name == "70"
+    # Kata contract (reference device_plugin.go:62-68)
+    assert d0.annotations["attach-pci"] == "true"
+    assert d0.annotations["bdf"] == "0000:0a:00.0"
+    assert d0.annotations["cdi.k8s.io/vfio70"] == "amd.com/gpu=70"
+    assert d0.device_nodes == [os.path.join(mock_cfg.dev_root, "vfio", "70")]
+
+
+def test_multifunction_group_one_device_one_node(tmp_path):
+    """Reference quirk NOT reproduced: per-function devices duplicated the
+    /dev/vfio node (device_plugin.go:59-77). Here: one device per group,
+    bdf annotation carries both functions."""
+    node = make_mock_node(str(tmp_path), n_gpus=1, with_audio_fn=True, kfd=False, hint=False)
+    cfg = node.config()
+    spec = build_spec(scan_node(cfg), "amd.com/gpu", cfg.dev_root)
+    assert len(spec.devices) == 1
+    d = spec.devices[0]
+    assert d.annotations["bdf"] == "0000:0a:00.0,0000:0a:00.1"
+    assert len(d.device_nodes) == 1
+
+
+def test_write_yaml_golden(mock_cfg, tmp_path):
+    inv = scan_node(mock_cfg)
+    spec = build_spec(inv, "amd.com/gpu", "/dev")
+    path = write_spec(spec, str(tmp_path), "kxdp-vfio", "yaml")
+    assert path.endswith("kxdp-vfio.yaml")
+    obj = yaml.safe_load(open(path))
+    assert obj["cdiVersion"] == "0.8.0"
+    assert obj["kind"] == "amd.com/gpu"
+    assert [d["name"] for d in obj["devices"]] == [str(70 + i) for i in range(8)]
+    dev = obj["devices"][3]
+    assert dev["annotations"]["attach-pci"] == "true"
+    assert dev["annotations"]["bdf"] == "0000:22:00.0"
+    assert dev["containerEdits"]["deviceNodes"] == [
+        {"path": "/dev/vfio/73", "permissions": "rw"}
+    ]
+    # file mode is world-readable (runtime reads it)
+    assert oct(os.stat(path).st_mode & 0o777) == "0o644"
+
+
+def test_write_json_and_format_switch_removes_stale(mock_cfg, tmp_path):
+    inv = scan_node(mock_cfg)
+    spec = build_spec(inv, "amd.com/gpu", "/dev")
+    ypath = write_spec(spec, str(tmp_path), "kxdp-vfio", "yaml")
+    jpath = write_spec(spec, str(tmp_path), "kxdp-vfio", "json")
+    assert os.path.exists(jpath)
+    assert not os.path.exists(ypath), "stale yaml spec must be removed"
+    back = read_spec(jpath)
+    assert back.device_names() == spec.device_names()
+    assert back.devices[0].annotations == spec.devices[0].annotations
+
+
+def test_roundtrip_yaml(mock_cfg, tmp_path):
+    inv = scan_node(mock_cfg)
+    spec = build_spec(inv, "amd.com/gpu", mock_cfg.dev_root)
+    path = write_spec(spec, str(tmp_path), "s", "yaml")
+    back = read_spec(path)
+    assert back.kind == spec.kind
+    assert back.device_names() == spec.device_names()
+    assert back.devices[5].device_nodes == spec.devices[5].device_nodes
+
+
+def test_validation_rejects_duplicates_and_empty(mock_cfg):
+    inv = scan_node(mock_cfg)
+    spec = build_spec(inv, "amd.com/gpu", mock_cfg.dev_root)
+    spec.devices.append(spec.devices[0])
+    with pytest.raises(ValueError, match="duplicate"):
+        spec.validate()
+    spec.devices.pop()
+    spec.devices[0].device_nodes = []
+    with pytest.raises(ValueError, match="no device nodes"):
+        spec.validate()
+
+
+def test_spec_path_helper(tmp_path):
+    assert spec_path(str(tmp_path), "n", "yaml").endswith("n.yaml")
+    assert spec_path(str(tmp_path), "n", "json").endswith("n.json")
