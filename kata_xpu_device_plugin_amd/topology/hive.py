@@ -191,7 +191,26 @@ def preferred_sets(
 
     affinities = [bucket_affinity(k) for k in keys]
 
-    best_score = -1
+    # Packing tie-break: among equal-locality choices prefer the set that
+    # leaves the least fragmentation — deplete small/partial hives first so
+    # a later large pod still finds an intact hive (config #4's 4+4 split).
+    def packing_key(key: Tuple[str, int]) -> Tuple[str, int]:
+        hive, numa = key
+        return (hive, -1) if hive else ("numa", numa)
+
+    hive_free: Dict[Tuple[str, int], int] = {}
+    for k, cap in zip(keys, caps):
+        hive_free[packing_key(k)] = hive_free.get(packing_key(k), 0) + cap
+
+    def packing(take: Sequence[int]) -> int:
+        taken: Dict[Tuple[str, int], int] = {}
+        for k, c in zip(keys, take):
+            if c:
+                taken[packing_key(k)] = taken.get(packing_key(k), 0) + c
+        # negative leftover across touched hives; 0 is best (hive depleted)
+        return -sum(hive_free[h] - c for h, c in taken.items())
+
+    best_score = (-1, 0)
     best_take: Optional[Tuple[int, ...]] = None
 
     def bucket_pair_score(key: Tuple[str, int], c: int) -> int:
@@ -228,7 +247,7 @@ def preferred_sets(
     def dfs(i: int, left: int, acc: int):
         nonlocal best_score, best_take
         if left == 0:
-            total = acc + cross_score(take)
+            total = (acc + cross_score(take), packing(take))
             if total > best_score:
                 best_score = total
                 best_take = tuple(take)
