@@ -1,0 +1,85 @@
+"""Prometheus metrics exporter.
+
+The reference has no observability beyond log.Printf (SURVEY.md §5 —
+prometheus libs are indirect deps only, never wired). Here: a real
+/metrics endpoint with allocation counters/latency, device health and
+discovery stats, enabled with --metrics-port.
+"""
+from __future__ import annotations
+
+from typing import Optional
+
+from prometheus_client import CollectorRegistry, Gauge, start_http_server
+
+from .utils.log import get_logger
+
+log = get_logger(__name__)
+
+
+class MetricsExporter:
+    def __init__(self, manager):
+        self.manager = manager
+        self.registry = CollectorRegistry()
+        self._server = None
+
+        self.g_devices = Gauge(
+            "kxdp_devices", "Discovered schedulable xPU devices",
+            ["resource"], registry=self.registry,
+        )
+        self.g_healthy = Gauge(
+            "kxdp_devices_healthy", "Healthy xPU devices",
+            ["resource"], registry=self.registry,
+        )
+        self.g_allocations = Gauge(
+            "kxdp_allocations_total", "Allocate() calls served",
+            ["resource"], registry=self.registry,
+        )
+        self.g_alloc_failures = Gauge(
+            "kxdp_allocation_failures_total", "Allocate() calls rejected",
+            ["resource"], registry=self.registry,
+        )
+        self.g_last_alloc = Gauge(
+            "kxdp_last_allocate_seconds", "Duration of the last Allocate()",
+            ["resource"], registry=self.registry,
+        )
+        self.g_scan_wall = Gauge(
+            "kxdp_discovery_seconds", "Wall time of the last discovery scan",
+            registry=self.registry,
+        )
+        # pull-model: refresh gauges right before each scrape
+        self.registry.register(_Refresher(self))
+
+    def refresh(self) -> None:
+        m = self.manager
+        if m.inventory is not None:
+            self.g_scan_wall.set(m.inventory.scan_wall_s)
+        for rname, plugin in m.plugins.items():
+            state = m.states[rname]
+            self.g_devices.labels(rname).set(len(state.device_ids()))
+            self.g_healthy.labels(rname).set(len(state.healthy_ids()))
+            self.g_allocations.labels(rname).set(plugin.allocations)
+            self.g_alloc_failures.labels(rname).set(plugin.allocate_failures)
+            self.g_last_alloc.labels(rname).set(plugin.last_allocate_s)
+
+    def start(self, port: int) -> None:
+        self._server, _ = start_http_server(port, registry=self.registry)
+        log.info("metrics on :%d/metrics", port)
+
+    def stop(self) -> None:
+        if self._server is not None:
+            self._server.shutdown()
+            self._server = None
+
+
+class _Refresher:
+    """Collector that refreshes gauges before each scrape."""
+
+    def __init__(self, exporter: MetricsExporter):
+        self.exporter = exporter
+
+    def collect(self):
+        try:
+            self.exporter.refresh()
+        except Exception:
+            log.exception("metrics refresh failed")
+        return []

@@ -1,0 +1,183 @@
+"""Lifecycle orchestrator: discovery → topology → CDI → plugins → watch.
+
+Reference analog: ``InitiateDevicePlugin`` + ``createDevicePlugins``
+(`pkg/device_plugin/device_plugin.go:44-53, :83-119`) which block forever
+on a never-closed channel with no signal handling (`:114` — SURVEY.md §7
+quirk list). This manager owns a proper shutdown path (SIGTERM/SIGINT),
+a single health watcher, and re-registration on kubelet restart.
+"""
+from __future__ import annotations
+
+import signal
+import threading
+from typing import Dict, List, Optional
+
+from ..cdi.spec import build_spec, write_spec
+from ..config import Config
+from ..discovery.naming import resource_name
+from ..discovery.sysfs import NodeInventory, XPUDevice, scan_node
+from ..health.watcher import NodeWatcher
+from ..topology.hive import GPUTopology, load_topology
+from ..utils.log import configure as configure_logging, get_logger
+from .server import XPUDevicePlugin
+from .state import DeviceState
+
+log = get_logger(__name__)
+
+
+class PluginManager:
+    def __init__(self, cfg: Config):
+        cfg.validate()
+        self.cfg = cfg
+        self.inventory: Optional[NodeInventory] = None
+        self.topology: Optional[GPUTopology] = None
+        self.plugins: Dict[str, XPUDevicePlugin] = {}   # resource name → plugin
+        self.states: Dict[str, DeviceState] = {}
+        self.watcher: Optional[NodeWatcher] = None
+        self.cdi_spec_path: Optional[str] = None
+        self._stop = threading.Event()
+        self._metrics = None
+
+    # ------------------------------------------------------------------
+    def _group_by_resource(self, inv: NodeInventory) -> Dict[str, Dict[str, XPUDevice]]:
+        out: Dict[str, Dict[str, XPUDevice]] = {}
+        for gid, dev in inv.devices.items():
+            rname = resource_name(
+                dev.model_device_id,
+                namespace=self.cfg.resource_namespace,
+                pci_ids_paths=tuple(self.cfg.pci_ids_paths),
+                is_vf=dev.is_vf,
+                unified=self.cfg.unified_resource_name,
+            )
+            out.setdefault(rname, {})[gid] = dev
+        return out
+
+    def setup(self) -> None:
+        """Discovery pass → CDI spec → per-resource plugin objects."""
+        inv = scan_node(self.cfg)
+        self.inventory = inv
+        log.info(
+            "discovered %d xPU device(s) in %.1f ms (%d %s function(s) total)",
+            len(inv.devices), inv.scan_wall_s * 1e3, len(inv.all_functions),
+            "/".join(f"{v:04x}" for v in self.cfg.vendor_allowlist),
+        )
+        self.topology = load_topology(self.cfg, inv)
+        log.info("xGMI topology source: %s (%d hive-mapped GPUs)",
+                 self.topology.source, len(self.topology.hive_of))
+
+        spec = build_spec(inv, self.cfg.cdi_kind, self.cfg.dev_root)
+        self.cdi_spec_path = write_spec(
+            spec, self.cfg.cdi_dir, self.cfg.cdi_spec_name, self.cfg.cdi_format
+        )
+
+        for rname, devs in sorted(self._group_by_resource(inv).items()):
+            state = DeviceState(devs)
+            self.states[rname] = state
+            self.plugins[rname] = XPUDevicePlugin(
+                self.cfg, rname, state, self.topology
+            )
+            log.info("resource %s: %d device(s): %s", rname, len(devs),
+                     ",".join(sorted(devs)))
+
+    # ------------------------------------------------------------------
+    def start(self, register: bool = True) -> None:
+        if not self.plugins:
+            log.warning("no devices discovered; serving nothing (will still "
+                        "watch for kubelet restarts)")
+        for plugin in self.plugins.values():
+            plugin.start(register=register)
+        self.watcher = NodeWatcher(
+            self.cfg,
+            self.states,
+            on_socket_removed=self._on_socket_removed,
+            on_kubelet_restarted=self._on_kubelet_restarted,
+            plugin_socket_names={p.socket_name for p in self.plugins.values()},
+        )
+        self.watcher.start()
+        self.watcher.wait_ready()
+        if self.cfg.metrics_port:
+            from ..metrics import MetricsExporter
+            self._metrics = MetricsExporter(self)
+            self._metrics.start(self.cfg.metrics_port)
+
+    def _on_socket_removed(self, socket_name: str) -> None:
+        for plugin in self.plugins.values():
+            if plugin.socket_name == socket_name and not self._stop.is_set():
+                try:
+                    plugin.restart()
+                except Exception:
+                    log.exception("restart of %s failed", plugin.resource_name)
+
+    def _on_kubelet_restarted(self) -> None:
+        if self._stop.is_set():
+            return
+        for plugin in self.plugins.values():
+            try:
+                plugin.register_with_kubelet()
+            except Exception:
+                log.exception("re-registration of %s failed", plugin.resource_name)
+
+    # ------------------------------------------------------------------
+    def run_forever(self) -> None:
+        """Block until SIGTERM/SIGINT (the reference blocks on a channel it
+        never closes and installs no signal handler)."""
+        def _sig(signum, frame):
+            log.info("signal %d — shutting down", signum)
+            self._stop.set()
+
+        signal.signal(signal.SIGTERM, _sig)
+        signal.signal(signal.SIGINT, _sig)
+        self._stop.wait()
+        self.stop()
+
+    def stop(self) -> None:
+        self._stop.set()
+        if self.watcher is not None:
+            self.watcher.stop()
+            self.watcher = None
+        for plugin in self.plugins.values():
+            plugin.stop()
+        if self._metrics is not None:
+            self._metrics.stop()
+            self._metrics = None
+
+    # ------------------------------------------------------------------
+    def rescan(self) -> None:
+        """Re-run discovery and swap device sets in place (e.g. after VF
+        count changes); plugins keep serving."""
+        inv = scan_node(self.cfg)
+        self.inventory = inv
+        spec = build_spec(inv, self.cfg.cdi_kind, self.cfg.dev_root)
+        self.cdi_spec_path = write_spec(
+            spec, self.cfg.cdi_dir, self.cfg.cdi_spec_name, self.cfg.cdi_format
+        )
+        grouped = self._group_by_resource(inv)
+        for rname, state in self.states.items():
+            state.replace_devices(grouped.get(rname, {}))
+        # brand-new resource names require a daemon restart (documented)
+        new = set(grouped) - set(self.states)
+        if new:
+            log.warning("rescan found new resource name(s) %s — restart the "
+                        "daemon to serve them", sorted(new))
+
+
+def main(argv: Optional[List[str]] = None) -> int:
+    import argparse
+
+    parser = argparse.ArgumentParser(
+        prog="kata-xpu-device-plugin-amd",
+        description="MI355X-native Kubernetes device plugin for VFIO/Kata GPU passthrough",
+    )
+    Config.add_args(parser)
+    args = parser.parse_args(argv)
+    cfg = Config.from_args(args)
+    configure_logging(cfg.log_level)
+    mgr = PluginManager(cfg)
+    mgr.setup()
+    mgr.start()
+    mgr.run_forever()
+    return 0
+
+
+if __name__ == "__main__":
+    raise SystemExit(main())

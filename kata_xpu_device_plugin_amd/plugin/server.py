@@ -1,0 +1,283 @@
+"""Per-resource DevicePlugin gRPC server.
+
+MI355X-native counterpart of the reference's ``GenericDevicePlugin``
+(`pkg/device_plugin/generic_device_plugin.go:37-50`, RPCs at :222, :320,
+:365, :372, :378). One server per resource name (one per GPU model, like
+the reference's one-per-deviceMap-entry at `device_plugin.go:83-119`).
+
+Differences from the reference, by design:
+
+* ``GetPreferredAllocation`` is implemented (xGMI-hive placement,
+  topology/hive.py) and advertised in GetDevicePluginOptions — the
+  reference returns nil,nil and does not advertise it.
+* ``Allocate`` supports three device-list strategies (cdi-cri default,
+  cdi-annotations, raw device-nodes) and does not clobber the env map the
+  response already carries (reference overwrites Envs at
+  `generic_device_plugin.go:348`).
+* ListAndWatch pushes immutable snapshots from DeviceState (no shared
+  mutable ``devs`` slice) and carries NUMA TopologyInfo so kubelet's
+  TopologyManager can align CPU/NIC placement.
+* Allocate re-validates each requested group against live sysfs like the
+  reference (`generic_device_plugin.go:329-338`) — via the native C++
+  scanner when available.
+"""
+from __future__ import annotations
+
+import os
+import queue
+import threading
+import time
+from concurrent import futures
+from typing import Dict, List, Optional
+
+import grpc
+
+from ..cdi.spec import ANNOTATION_PREFIX, qualified_name
+from ..config import (
+    Config,
+    STRATEGY_CDI_ANNOTATIONS,
+    STRATEGY_CDI_CRI,
+    STRATEGY_DEVICE_NODES,
+)
+from ..discovery.sysfs import read_hex, read_link_base
+from ..topology.hive import GPUTopology, preferred_sets
+from ..utils.log import get_logger
+from . import api
+from .state import DeviceState
+
+log = get_logger(__name__)
+
+# Env var consumed by the Kata-side runtime to locate the CDI vendor class
+# (reference contract: generic_device_plugin.go:30-31, :348-350).
+ENV_CDI_VENDOR_CLASS = "KUBERNETES_CDI_VENDOR_CLASS"
+# Informational env listing the passed-through PCI addresses, in the spirit
+# of the PCI_RESOURCE_* convention (reference buildEnv,
+# generic_device_plugin.go:90-96 — dead there, live here).
+ENV_PCI_RESOURCE_PREFIX = "PCI_RESOURCE_"
+
+
+class AllocationError(Exception):
+    pass
+
+
+class XPUDevicePlugin:
+    """gRPC servicer + server lifecycle for one resource name."""
+
+    def __init__(
+        self,
+        cfg: Config,
+        resource_name: str,
+        state: DeviceState,
+        topo: GPUTopology,
+        socket_name: Optional[str] = None,
+    ):
+        self.cfg = cfg
+        self.resource_name = resource_name
+        self.state = state
+        self.topo = topo
+        short = resource_name.split("/", 1)[1].lower()
+        self.socket_name = socket_name or f"{cfg.plugin_socket_prefix}-{short}.sock"
+        self.socket_path = os.path.join(cfg.kubelet_socket_dir, self.socket_name)
+        self._server: Optional[grpc.Server] = None
+        self._stop = threading.Event()
+        self._serving = threading.Event()
+        self.allocations = 0          # metrics
+        self.allocate_failures = 0
+        self.last_allocate_s = 0.0
+
+    # ------------------------------------------------------------------
+    # DevicePlugin service
+    # ------------------------------------------------------------------
+    def GetDevicePluginOptions(self, request, context):
+        return api.DevicePluginOptions(
+            pre_start_required=False,
+            get_preferred_allocation_available=True,
+        )
+
+    def _device_list(self) -> List[object]:
+        devs = []
+        for dev, healthy in self.state.snapshot():
+            d = api.Device(
+                id=dev.id,
+                health=api.HEALTHY if healthy else api.UNHEALTHY,
+            )
+            if dev.numa_node >= 0:
+                d.topology.nodes.add(id=dev.numa_node)
+            devs.append(d)
+        return devs
+
+    def ListAndWatch(self, request, context):
+        """Initial full list, then a push per health/inventory change
+        (reference: generic_device_plugin.go:222-250)."""
+        q = self.state.watch()
+        try:
+            yield api.ListAndWatchResponse(devices=self._device_list())
+            while not self._stop.is_set() and context.is_active():
+                try:
+                    q.get(timeout=0.5)
+                except queue.Empty:
+                    continue
+                # coalesce bursts of updates into one push
+                while True:
+                    try:
+                        q.get_nowait()
+                    except queue.Empty:
+                        break
+                yield api.ListAndWatchResponse(devices=self._device_list())
+        finally:
+            self.state.unwatch(q)
+
+    # -- Allocate ------------------------------------------------------
+    def _revalidate(self, gid: str) -> None:
+        """Check the group still exists on this node with our vendor bound
+        to vfio (reference re-reads iommu_group+vendor per function:
+        generic_device_plugin.go:329-338)."""
+        dev = self.state.device(gid)
+        if dev is None:
+            raise AllocationError(f"unknown device id {gid}")
+        devices_dir = os.path.join(self.cfg.sysfs_root, "bus", "pci", "devices")
+        for fn in dev.functions:
+            p = os.path.join(devices_dir, fn.bdf)
+            if read_link_base(os.path.join(p, "iommu_group")) != gid:
+                raise AllocationError(f"device {fn.bdf} no longer in IOMMU group {gid}")
+            vendor = read_hex(os.path.join(p, "vendor"))
+            if vendor not in set(self.cfg.vendor_allowlist):
+                raise AllocationError(f"device {fn.bdf} vendor {vendor:#x} not allowed")
+            if read_link_base(os.path.join(p, "driver")) != self.cfg.required_driver:
+                raise AllocationError(
+                    f"device {fn.bdf} not bound to {self.cfg.required_driver}"
+                )
+
+    def _container_response(self, ids: List[str]):
+        resp = api.ContainerAllocateResponse()
+        strategy = self.cfg.device_list_strategy
+        bdfs: List[str] = []
+        for gid in ids:
+            dev = self.state.device(gid)
+            assert dev is not None  # validated earlier
+            bdfs.extend(dev.bdfs)
+            if strategy == STRATEGY_CDI_CRI:
+                resp.cdi_devices.add(name=qualified_name(self.cfg.cdi_kind, gid))
+            elif strategy == STRATEGY_CDI_ANNOTATIONS:
+                key = f"{ANNOTATION_PREFIX}vfio{gid}"
+                resp.annotations[key] = qualified_name(self.cfg.cdi_kind, gid)
+            else:  # raw device nodes, no CDI
+                path = os.path.join(self.cfg.dev_root, dev.vfio_node)
+                resp.devices.add(
+                    container_path=path, host_path=path, permissions="rw"
+                )
+        if strategy in (STRATEGY_CDI_CRI, STRATEGY_CDI_ANNOTATIONS):
+            resp.envs[ENV_CDI_VENDOR_CLASS] = self.cfg.cdi_kind
+        env_res = ENV_PCI_RESOURCE_PREFIX + self.resource_name.upper().replace(
+            "/", "_").replace(".", "_").replace("-", "_")
+        resp.envs[env_res] = ",".join(bdfs)
+        return resp
+
+    def Allocate(self, request, context):
+        """The hot path (BASELINE metric). Reference:
+        generic_device_plugin.go:320-355."""
+        t0 = time.perf_counter()
+        response = api.AllocateResponse()
+        try:
+            for creq in request.container_requests:
+                ids = list(creq.devices_ids)
+                for gid in ids:
+                    self._revalidate(gid)
+                response.container_responses.append(self._container_response(ids))
+        except AllocationError as e:
+            self.allocate_failures += 1
+            log.warning("Allocate(%s) rejected: %s", self.resource_name, e)
+            context.abort(grpc.StatusCode.INVALID_ARGUMENT, str(e))
+        self.allocations += 1
+        self.last_allocate_s = time.perf_counter() - t0
+        return response
+
+    def GetPreferredAllocation(self, request, context):
+        resp = api.PreferredAllocationResponse()
+        bdf_of: Dict[str, str] = {}
+        for gid in self.state.device_ids():
+            dev = self.state.device(gid)
+            if dev:
+                bdf_of[gid] = dev.primary.bdf
+        for creq in request.container_requests:
+            available = [d for d in creq.available_device_ids if d in bdf_of]
+            must = list(creq.must_include_device_ids)
+            pick = preferred_sets(
+                self.topo, bdf_of, available, must, creq.allocation_size
+            )
+            resp.container_responses.add(device_ids=pick)
+        return resp
+
+    def PreStartContainer(self, request, context):
+        return api.PreStartContainerResponse()
+
+    # ------------------------------------------------------------------
+    # lifecycle (reference: Start/Stop/Register,
+    # generic_device_plugin.go:128-219)
+    # ------------------------------------------------------------------
+    def start(self, register: bool = True) -> None:
+        self._stop.clear()
+        server = grpc.server(
+            futures.ThreadPoolExecutor(max_workers=8),
+            options=[("grpc.max_concurrent_streams", 64)],
+        )
+        api.add_device_plugin_servicer(server, self)
+        if os.path.exists(self.socket_path):
+            os.unlink(self.socket_path)
+        server.add_insecure_port(f"unix://{self.socket_path}")
+        server.start()
+        self._server = server
+        # self-dial until ready (reference waitForGrpcServer :98-105)
+        ch = grpc.insecure_channel(f"unix://{self.socket_path}")
+        try:
+            grpc.channel_ready_future(ch).result(timeout=self.cfg.grpc_timeout_s)
+        finally:
+            ch.close()
+        self._serving.set()
+        if register:
+            self.register_with_kubelet()
+        log.info("plugin %s serving on %s", self.resource_name, self.socket_path)
+
+    def register_with_kubelet(self) -> None:
+        """Register against kubelet.sock (reference Register :200-219)."""
+        ch = grpc.insecure_channel(f"unix://{self.cfg.kubelet_socket}")
+        try:
+            grpc.channel_ready_future(ch).result(timeout=self.cfg.grpc_timeout_s)
+            stub = api.RegistrationStub(ch)
+            stub.Register(
+                api.RegisterRequest(
+                    version=api.VERSION,
+                    endpoint=self.socket_name,
+                    resource_name=self.resource_name,
+                    options=api.DevicePluginOptions(
+                        pre_start_required=False,
+                        get_preferred_allocation_available=True,
+                    ),
+                ),
+                timeout=self.cfg.grpc_timeout_s,
+            )
+        finally:
+            ch.close()
+
+    def stop(self) -> None:
+        self._stop.set()
+        self._serving.clear()
+        if self._server is not None:
+            self._server.stop(grace=1.0).wait()
+            self._server = None
+        if os.path.exists(self.socket_path):
+            try:
+                os.unlink(self.socket_path)
+            except OSError:
+                pass
+
+    def restart(self, register: bool = True) -> None:
+        """Full stop/start cycle, e.g. after kubelet restart (reference
+        restart :186-197 — whose fresh local stop channel detached the
+        restarted plugin from global shutdown; our single Event does not)."""
+        self.stop()
+        self.start(register=register)
+
+    @property
+    def serving(self) -> bool:
+        return self._serving.is_set()

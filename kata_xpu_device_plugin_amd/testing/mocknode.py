@@ -20,6 +20,7 @@ Default shape: one 8×MI355X OAM node, all 8 GPUs in one xGMI hive with
 from __future__ import annotations
 
 import json
+import tempfile
 import os
 from dataclasses import dataclass, field
 from typing import Dict, List, Optional, Sequence
@@ -65,11 +66,14 @@ class MockNode:
         return os.path.join(self.root, "dev")
 
     def config(self, **overrides) -> Config:
+        # Unix socket paths are capped at ~107 chars; pytest tmp roots are
+        # deep, so sockets live in a short mkdtemp instead of under root.
+        sock_dir = tempfile.mkdtemp(prefix="kxdp-")
         cfg = Config(
             sysfs_root=self.sysfs,
             dev_root=self.dev,
             cdi_dir=os.path.join(self.root, "var", "run", "cdi"),
-            kubelet_socket_dir=os.path.join(self.root, "kubelet", "device-plugins"),
+            kubelet_socket_dir=sock_dir,
             topology_hint_path=os.path.join(self.root, "etc", "topology.json"),
             pci_ids_paths=(),
             metrics_port=0,
