@@ -131,11 +131,26 @@ class XPUDevicePlugin:
     def _revalidate(self, gid: str) -> None:
         """Check the group still exists on this node with our vendor bound
         to vfio (reference re-reads iommu_group+vendor per function:
-        generic_device_plugin.go:329-338)."""
+        generic_device_plugin.go:329-338). Native C++ path when built."""
         dev = self.state.device(gid)
         if dev is None:
             raise AllocationError(f"unknown device id {gid}")
         devices_dir = os.path.join(self.cfg.sysfs_root, "bus", "pci", "devices")
+        if self.cfg.native != "off":
+            try:
+                from .. import _native
+            except ImportError:
+                _native = None
+                if self.cfg.native == "require":
+                    raise RuntimeError("_native extension required but missing")
+            if _native is not None:
+                err = _native.revalidate_group(
+                    devices_dir, gid, [fn.bdf for fn in dev.functions],
+                    list(self.cfg.vendor_allowlist), self.cfg.required_driver,
+                )
+                if err:
+                    raise AllocationError(err)
+                return
         for fn in dev.functions:
             p = os.path.join(devices_dir, fn.bdf)
             if read_link_base(os.path.join(p, "iommu_group")) != gid:

@@ -151,6 +151,7 @@ def preferred_sets(
     available: Sequence[str],
     must_include: Sequence[str],
     size: int,
+    use_native: bool = True,
 ) -> List[str]:
     """Choose `size` device IDs from `available` (⊇ must_include).
 
@@ -168,6 +169,17 @@ def preferred_sets(
             return []
     if len(must) >= size:
         return must[:size]
+
+    try:
+        from .. import _native
+    except ImportError:
+        _native = None
+    if use_native and _native is not None:
+        locality = {
+            d: (topo.hive(bdf_of[d]), topo.numa_of.get(bdf_of[d], -1))
+            for d in avail
+        }
+        return _native.select_preferred(locality, avail, must, size)
 
     remaining_ids = [d for d in avail if d not in set(must)]
     need = size - len(must)
