@@ -65,11 +65,13 @@ class Config:
     )
     # Only devices bound to this driver are schedulable (VFIO passthrough).
     required_driver: str = field(default_factory=lambda: _env("DRIVER", "vfio-pci"))
-    # Restrict to PCI class 0x03xxxx (display/3D). The reference filtered on
-    # vendor only (device_plugin.go:142-149), which would also match vendor
-    # audio/bridge functions; we keep non-GPU functions of the same IOMMU
-    # group as passthrough companions but never schedule them.
-    gpu_class_prefix: int = 0x03
+    # Schedulable PCI classes: 0x12xxxx "Processing accelerator" (MI355X
+    # enumerates as 0x120000 — live-node confirmed) and 0x03xxxx display
+    # (consumer AMD GPUs). The reference filtered on vendor only
+    # (device_plugin.go:142-149), which would also match vendor audio/bridge
+    # functions; we keep non-GPU functions of the same IOMMU group as
+    # passthrough companions but never schedule them.
+    gpu_class_prefixes: Sequence[int] = (0x03, 0x12)
 
     # --- naming ---
     resource_namespace: str = field(default_factory=lambda: _env("NAMESPACE", "amd.com"))

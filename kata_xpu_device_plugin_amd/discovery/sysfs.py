@@ -59,7 +59,12 @@ class PCIFunction:
 
     @property
     def is_gpu(self) -> bool:
-        return (self.class_code >> 16) == 0x03
+        # MI355X (and MI300-series) enumerate as PCI class 0x1200xx
+        # "Processing accelerator", NOT display class 0x03 — confirmed on a
+        # live 8×MI355X node (device 0x75a3, class 0x120000, one GPU per
+        # IOMMU group behind an AMD 0x1501 bridge). Consumer/display AMD
+        # GPUs are 0x03xxxx; accept both.
+        return (self.class_code >> 16) in (0x03, 0x12)
 
 
 @dataclass
@@ -236,8 +241,9 @@ def scan_node(cfg: Config) -> NodeInventory:
             continue
         groups.setdefault(fn.iommu_group, []).append(fn)
 
+    class_prefixes = set(cfg.gpu_class_prefixes)
     for gid, fns in groups.items():
-        gpus = [f for f in fns if f.is_gpu]
+        gpus = [f for f in fns if (f.class_code >> 16) in class_prefixes]
         if not gpus:
             continue  # group has no GPU function — not a schedulable xPU
         if device_allow and not any(f.device in device_allow for f in gpus):

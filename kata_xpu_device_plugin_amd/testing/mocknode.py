@@ -28,7 +28,7 @@ from typing import Dict, List, Optional, Sequence
 from ..config import Config
 
 MI355X_DEVICE_ID = 0x75A3
-GPU_CLASS = 0x038000       # Display controller, other
+GPU_CLASS = 0x120000       # Processing accelerator (real MI355X class, live-node confirmed)
 AUDIO_CLASS = 0x040300     # Audio device (companion function example)
 
 
