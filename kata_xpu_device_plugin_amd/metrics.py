@@ -21,6 +21,9 @@ class MetricsExporter:
         self.manager = manager
         self.registry = CollectorRegistry()
         self._server = None
+        # Registered FIRST so each scrape refreshes the gauges before the
+        # registry collects them (collectors run in registration order).
+        self.registry.register(_Refresher(self))
 
         self.g_devices = Gauge(
             "kxdp_devices", "Discovered schedulable xPU devices",
@@ -46,8 +49,6 @@ class MetricsExporter:
             "kxdp_discovery_seconds", "Wall time of the last discovery scan",
             registry=self.registry,
         )
-        # pull-model: refresh gauges right before each scrape
-        self.registry.register(_Refresher(self))
 
     def refresh(self) -> None:
         m = self.manager

@@ -37,6 +37,7 @@ class PluginManager:
         self.cdi_spec_path: Optional[str] = None
         self._stop = threading.Event()
         self._metrics = None
+        self._amdsmi = None
 
     # ------------------------------------------------------------------
     def _group_by_resource(self, inv: NodeInventory) -> Dict[str, Dict[str, XPUDevice]]:
@@ -95,6 +96,12 @@ class PluginManager:
         )
         self.watcher.start()
         self.watcher.wait_ready()
+        if self.cfg.amdsmi_health:
+            from ..health.amdsmi_health import AmdSmiPoller
+            self._amdsmi = AmdSmiPoller(
+                self.cfg.health_poll_interval_s, self._on_amdsmi_health
+            )
+            self._amdsmi.start()
         if self.cfg.metrics_port:
             from ..metrics import MetricsExporter
             self._metrics = MetricsExporter(self)
@@ -107,6 +114,19 @@ class PluginManager:
                     plugin.restart()
                 except Exception:
                     log.exception("restart of %s failed", plugin.resource_name)
+
+    def _on_amdsmi_health(self, bdf: str, healthy: bool, reasons) -> None:
+        """amd-smi verdict → DeviceState (matches any function BDF of a
+        schedulable group; normally only fires on hybrid/pre-flight nodes
+        since vfio-bound GPUs are invisible to amd-smi)."""
+        for state in self.states.values():
+            for gid in state.device_ids():
+                dev = state.device(gid)
+                if dev and bdf in (fn.bdf.lower() for fn in dev.functions):
+                    if not healthy:
+                        log.warning("amd-smi: %s unhealthy: %s", bdf, reasons)
+                    state.set_health(gid, healthy)
+                    return
 
     def _on_kubelet_restarted(self) -> None:
         if self._stop.is_set():
@@ -135,6 +155,9 @@ class PluginManager:
         if self.watcher is not None:
             self.watcher.stop()
             self.watcher = None
+        if self._amdsmi is not None:
+            self._amdsmi.stop()
+            self._amdsmi = None
         for plugin in self.plugins.values():
             plugin.stop()
         if self._metrics is not None:

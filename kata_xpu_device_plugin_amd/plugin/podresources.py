@@ -1,0 +1,147 @@
+"""kubelet pod-resources client — live, unlike the reference's dead code.
+
+The reference ships a pod-resources client that is never called
+(`utils/pod_resources.go:41-61`, SURVEY.md C11 "dead code"). Here it is
+wired for real: ``assignments()`` lists which pods/containers currently
+hold which of our device IDs (kubelet's
+/var/lib/kubelet/pod-resources/kubelet.sock, v1 PodResourcesLister API),
+surfaced through the metrics exporter and the
+``python -m kata_xpu_device_plugin_amd.tools.assignments`` CLI.
+
+The v1 protobuf schema is reconstructed the same way as plugin/api.py
+(no protoc in the environment); field numbers follow
+k8s.io/kubelet/pkg/apis/podresources/v1/api.proto and are pinned by
+wire-format tests.
+"""
+from __future__ import annotations
+
+import os
+from dataclasses import dataclass, field
+from typing import Dict, List
+
+import grpc
+from google.protobuf import descriptor_pb2, descriptor_pool, message_factory
+
+_PKG = "v1podres"
+_F = descriptor_pb2.FieldDescriptorProto
+
+DEFAULT_SOCKET = "/var/lib/kubelet/pod-resources/kubelet.sock"
+
+
+def _build_fdp() -> descriptor_pb2.FileDescriptorProto:
+    fdp = descriptor_pb2.FileDescriptorProto()
+    fdp.name = "kxdp/podresources_v1.proto"
+    fdp.package = _PKG
+    fdp.syntax = "proto3"
+
+    def msg(name):
+        m = fdp.message_type.add()
+        m.name = name
+        return m
+
+    def fld(m, name, number, ftype, label=_F.LABEL_OPTIONAL, type_name=None):
+        f = m.field.add()
+        f.name, f.number, f.type, f.label = name, number, ftype, label
+        if type_name:
+            f.type_name = f".{_PKG}.{type_name}"
+
+    msg("ListPodResourcesRequest")
+
+    m = msg("ContainerDevices")
+    fld(m, "resource_name", 1, _F.TYPE_STRING)
+    fld(m, "device_ids", 2, _F.TYPE_STRING, _F.LABEL_REPEATED)
+    # field 3 is the TopologyInfo; not needed for assignment listing
+
+    m = msg("ContainerResources")
+    fld(m, "name", 1, _F.TYPE_STRING)
+    fld(m, "devices", 2, _F.TYPE_MESSAGE, _F.LABEL_REPEATED, "ContainerDevices")
+
+    m = msg("PodResources")
+    fld(m, "name", 1, _F.TYPE_STRING)
+    fld(m, "namespace", 2, _F.TYPE_STRING)
+    fld(m, "containers", 3, _F.TYPE_MESSAGE, _F.LABEL_REPEATED, "ContainerResources")
+
+    m = msg("ListPodResourcesResponse")
+    fld(m, "pod_resources", 1, _F.TYPE_MESSAGE, _F.LABEL_REPEATED, "PodResources")
+    return fdp
+
+
+_pool = descriptor_pool.DescriptorPool()
+_pool.Add(_build_fdp())
+
+
+def _cls(name):
+    return message_factory.GetMessageClass(_pool.FindMessageTypeByName(f"{_PKG}.{name}"))
+
+
+ListPodResourcesRequest = _cls("ListPodResourcesRequest")
+ContainerDevices = _cls("ContainerDevices")
+ContainerResources = _cls("ContainerResources")
+PodResources = _cls("PodResources")
+ListPodResourcesResponse = _cls("ListPodResourcesResponse")
+
+_SERVICE = f"{_PKG}.PodResourcesLister"
+# kubelet serves "v1.PodResourcesLister"; our package name must match the
+# wire path, so the real client overrides the service path:
+KUBELET_SERVICE_PATH = "/v1.PodResourcesLister/List"
+
+
+@dataclass
+class Assignment:
+    namespace: str
+    pod: str
+    container: str
+    resource_name: str
+    device_ids: List[str] = field(default_factory=list)
+
+
+class PodResourcesClient:
+    def __init__(self, socket_path: str = DEFAULT_SOCKET, timeout_s: float = 5.0):
+        self.socket_path = socket_path
+        self.timeout_s = timeout_s
+
+    def list(self) -> List[Assignment]:
+        ch = grpc.insecure_channel(f"unix://{self.socket_path}")
+        try:
+            grpc.channel_ready_future(ch).result(timeout=self.timeout_s)
+            call = ch.unary_unary(
+                KUBELET_SERVICE_PATH,
+                request_serializer=ListPodResourcesRequest.SerializeToString,
+                response_deserializer=ListPodResourcesResponse.FromString,
+            )
+            resp = call(ListPodResourcesRequest(), timeout=self.timeout_s)
+        finally:
+            ch.close()
+        out: List[Assignment] = []
+        for pod in resp.pod_resources:
+            for ctr in pod.containers:
+                for dev in ctr.devices:
+                    out.append(Assignment(
+                        namespace=pod.namespace, pod=pod.name,
+                        container=ctr.name, resource_name=dev.resource_name,
+                        device_ids=list(dev.device_ids),
+                    ))
+        return out
+
+    def assignments(self, resource_prefix: str = "amd.com/") -> Dict[str, str]:
+        """device id → 'namespace/pod/container' for our resources."""
+        out: Dict[str, str] = {}
+        for a in self.list():
+            if a.resource_name.startswith(resource_prefix):
+                for did in a.device_ids:
+                    out[did] = f"{a.namespace}/{a.pod}/{a.container}"
+        return out
+
+
+def add_lister_servicer(server: grpc.Server, servicer, service_name: str = "v1.PodResourcesLister") -> None:
+    """Server-side wiring (used by the test kubelet stub)."""
+    handlers = {
+        "List": grpc.unary_unary_rpc_method_handler(
+            servicer.List,
+            request_deserializer=ListPodResourcesRequest.FromString,
+            response_serializer=ListPodResourcesResponse.SerializeToString,
+        ),
+    }
+    server.add_generic_rpc_handlers(
+        (grpc.method_handlers_generic_handler(service_name, handlers),)
+    )

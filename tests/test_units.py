@@ -1,0 +1,263 @@
+"""Unit tests: config, DeviceState, inotify, amd-smi poller, metrics,
+allocate strategies, manager rescan."""
+import os
+import threading
+import time
+
+import pytest
+
+from kata_xpu_device_plugin_amd.config import Config
+from kata_xpu_device_plugin_amd.discovery import scan_node
+from kata_xpu_device_plugin_amd.health.amdsmi_health import AmdSmiPoller, DeviceHealth
+from kata_xpu_device_plugin_amd.plugin import api
+from kata_xpu_device_plugin_amd.plugin.manager import PluginManager
+from kata_xpu_device_plugin_amd.plugin.state import DeviceState
+from kata_xpu_device_plugin_amd.testing.kubelet_stub import KubeletStub
+from kata_xpu_device_plugin_amd.testing.mocknode import MockGPU, make_mock_node
+from kata_xpu_device_plugin_amd.utils import inotify
+
+
+# --- config ----------------------------------------------------------------
+
+def test_config_env_parsing(monkeypatch):
+    monkeypatch.setenv("KXDP_VENDORS", "1002,10ee")
+    monkeypatch.setenv("KXDP_DEVICES", "75a3")
+    monkeypatch.setenv("KXDP_STRATEGY", "cdi-annotations")
+    monkeypatch.setenv("KXDP_METRICS_PORT", "9110")
+    monkeypatch.setenv("KXDP_AMDSMI_HEALTH", "false")
+    cfg = Config()
+    cfg.validate()
+    assert cfg.vendor_allowlist == (0x1002, 0x10EE)
+    assert cfg.device_allowlist == (0x75A3,)
+    assert cfg.device_list_strategy == "cdi-annotations"
+    assert cfg.metrics_port == 9110
+    assert cfg.amdsmi_health is False
+
+
+def test_config_validation_errors():
+    cfg = Config(device_list_strategy="bogus")
+    with pytest.raises(ValueError):
+        cfg.validate()
+    cfg = Config(cdi_kind="nokind")
+    with pytest.raises(ValueError):
+        cfg.validate()
+    cfg = Config(cdi_format="xml")
+    with pytest.raises(ValueError):
+        cfg.validate()
+
+
+def test_config_from_args():
+    import argparse
+    p = argparse.ArgumentParser()
+    Config.add_args(p)
+    args = p.parse_args(["--strategy", "device-nodes", "--namespace", "x.org"])
+    cfg = Config.from_args(args)
+    assert cfg.device_list_strategy == "device-nodes"
+    assert cfg.resource_namespace == "x.org"
+
+
+# --- DeviceState -----------------------------------------------------------
+
+def _state_of(tmp_path, n=2):
+    node = make_mock_node(str(tmp_path), n_gpus=n, kfd=False, hint=False)
+    inv = scan_node(node.config())
+    return DeviceState(inv.devices), inv
+
+
+def test_state_snapshot_and_health(tmp_path):
+    st, inv = _state_of(tmp_path)
+    assert st.device_ids() == ["70", "71"]
+    assert st.healthy_ids() == ["70", "71"]
+    assert st.set_health("70", False)
+    assert not st.set_health("70", False)  # no change → no notify
+    assert st.healthy_ids() == ["71"]
+    assert not st.set_health("nope", False)
+
+
+def test_state_watch_notification(tmp_path):
+    st, inv = _state_of(tmp_path)
+    q = st.watch()
+    st.set_health("70", False)
+    assert q.get(timeout=1) is not None
+    st.unwatch(q)
+    st.set_health("70", True)
+    assert q.empty()
+
+
+def test_state_replace_devices_preserves_health(tmp_path):
+    st, inv = _state_of(tmp_path, n=3)
+    st.set_health("71", False)
+    st.replace_devices({g: d for g, d in inv.devices.items() if g != "72"})
+    assert st.device_ids() == ["70", "71"]
+    assert st.healthy_ids() == ["70"]  # 71 stays unhealthy
+
+
+def test_state_concurrent_mutation(tmp_path):
+    st, inv = _state_of(tmp_path, n=8)
+    stop = threading.Event()
+    errs = []
+
+    def flipper(gid):
+        try:
+            while not stop.is_set():
+                st.set_health(gid, False)
+                st.set_health(gid, True)
+        except Exception as e:  # pragma: no cover
+            errs.append(e)
+
+    def reader():
+        try:
+            while not stop.is_set():
+                snap = st.snapshot()
+                assert len(snap) == 8
+        except Exception as e:  # pragma: no cover
+            errs.append(e)
+
+    ts = [threading.Thread(target=flipper, args=(str(70 + i),)) for i in range(4)]
+    ts += [threading.Thread(target=reader) for _ in range(2)]
+    for t in ts:
+        t.start()
+    time.sleep(0.3)
+    stop.set()
+    for t in ts:
+        t.join()
+    assert not errs
+
+
+# --- inotify ---------------------------------------------------------------
+
+def test_inotify_create_delete(tmp_path):
+    with inotify.Inotify() as ino:
+        ino.add_watch(str(tmp_path), inotify.IN_CREATE | inotify.IN_DELETE)
+        f = tmp_path / "x"
+        f.write_text("")
+        evs = ino.read_events(timeout=2)
+        assert any(e.name == "x" and e.created for e in evs)
+        f.unlink()
+        evs = ino.read_events(timeout=2)
+        assert any(e.name == "x" and e.removed for e in evs)
+
+
+def test_inotify_bad_path():
+    with inotify.Inotify() as ino:
+        with pytest.raises(OSError):
+            ino.add_watch("/does/not/exist", inotify.IN_CREATE)
+
+
+# --- amd-smi poller (faked snapshots) --------------------------------------
+
+def test_amdsmi_poller_transitions():
+    calls = []
+    healthy = {"0000:0a:00.0": True}
+
+    def snap():
+        return {b: DeviceHealth(bdf=b, healthy=h, reasons=[] if h else ["ecc"])
+                for b, h in healthy.items()}
+
+    p = AmdSmiPoller(999, calls_append := (lambda b, h, r: calls.append((b, h))),
+                     snapshot_fn=snap)
+    p.poll_once()
+    assert calls == []  # initially healthy → no event
+    healthy["0000:0a:00.0"] = False
+    p.poll_once()
+    assert calls == [("0000:0a:00.0", False)]
+    p.poll_once()
+    assert calls == [("0000:0a:00.0", False)]  # no repeat
+    healthy["0000:0a:00.0"] = True
+    p.poll_once()
+    assert calls[-1] == ("0000:0a:00.0", True)
+
+
+def test_amdsmi_initially_unhealthy_reported():
+    calls = []
+    p = AmdSmiPoller(
+        999, lambda b, h, r: calls.append((b, h)),
+        snapshot_fn=lambda: {"b": DeviceHealth(bdf="b", healthy=False,
+                                               reasons=["ecc"])},
+    )
+    p.poll_once()
+    assert calls == [("b", False)]
+
+
+def test_amdsmi_snapshot_no_library_is_empty_or_dict():
+    # On CPU boxes amdsmi may import but see no devices; must not raise.
+    from kata_xpu_device_plugin_amd.health.amdsmi_health import snapshot
+    assert isinstance(snapshot(), dict)
+
+
+# --- allocate strategies ---------------------------------------------------
+
+@pytest.mark.parametrize("strategy,check", [
+    ("cdi-cri", "cdi"),
+    ("cdi-annotations", "ann"),
+    ("device-nodes", "nodes"),
+])
+def test_allocate_strategies(tmp_path, strategy, check):
+    node = make_mock_node(str(tmp_path), n_gpus=2)
+    cfg = node.config(device_list_strategy=strategy)
+    stub = KubeletStub(cfg.kubelet_socket_dir)
+    stub.start()
+    mgr = PluginManager(cfg)
+    mgr.setup()
+    mgr.start()
+    try:
+        reg = stub.wait_for_registration(1)[0]
+        ps = stub.plugin_stub(reg.endpoint)
+        resp = ps.Allocate(api.AllocateRequest(container_requests=[
+            api.ContainerAllocateRequest(devices_ids=["70"]),
+        ]))
+        cr = resp.container_responses[0]
+        if check == "cdi":
+            assert [c.name for c in cr.cdi_devices] == ["amd.com/gpu=70"]
+            assert len(cr.devices) == 0 and len(cr.annotations) == 0
+        elif check == "ann":
+            assert cr.annotations["cdi.k8s.io/vfio70"] == "amd.com/gpu=70"
+            assert len(cr.cdi_devices) == 0
+        else:
+            assert cr.devices[0].host_path.endswith("/vfio/70")
+            assert cr.devices[0].permissions == "rw"
+            assert len(cr.cdi_devices) == 0
+            assert "KUBERNETES_CDI_VENDOR_CLASS" not in cr.envs
+    finally:
+        mgr.stop()
+        stub.stop()
+
+
+# --- manager rescan --------------------------------------------------------
+
+def test_manager_rescan_picks_up_new_device(tmp_path):
+    node = make_mock_node(str(tmp_path), n_gpus=2, kfd=False, hint=False)
+    cfg = node.config()
+    mgr = PluginManager(cfg)
+    mgr.setup()
+    assert len(mgr.inventory.devices) == 2
+    node.add_gpu(MockGPU(bdf="0000:66:00.0", iommu_group="99"))
+    mgr.rescan()
+    assert len(mgr.inventory.devices) == 3
+    state = mgr.states["amd.com/INSTINCT_MI355X"]
+    assert "99" in state.device_ids()
+
+
+def test_manager_unified_resource(tmp_path):
+    node = make_mock_node(str(tmp_path), n_gpus=2, kfd=False, hint=False)
+    node.add_gpu(MockGPU(bdf="0000:66:00.0", device_id=0x74A1, iommu_group="99"))
+    cfg = node.config(unified_resource_name="gpu")
+    mgr = PluginManager(cfg)
+    mgr.setup()
+    assert list(mgr.plugins) == ["amd.com/GPU"]
+    assert len(mgr.states["amd.com/GPU"].device_ids()) == 3
+
+
+# --- metrics ---------------------------------------------------------------
+
+def test_metrics_exporter_refresh(tmp_path):
+    from prometheus_client import generate_latest
+    from kata_xpu_device_plugin_amd.metrics import MetricsExporter
+
+    node = make_mock_node(str(tmp_path), n_gpus=2, kfd=False, hint=False)
+    mgr = PluginManager(node.config())
+    mgr.setup()
+    exp = MetricsExporter(mgr)
+    text = generate_latest(exp.registry).decode()
+    assert 'kxdp_devices{resource="amd.com/INSTINCT_MI355X"} 2.0' in text
+    assert "kxdp_discovery_seconds" in text
