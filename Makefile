@@ -1,0 +1,25 @@
+PYTHON ?= python3
+IMAGE  ?= ghcr.io/example/kata-xpu-device-plugin-amd:0.1.0
+
+.PHONY: build build-hip test test-gpu bench image clean
+
+build:
+	$(PYTHON) setup.py build_ext --inplace
+
+build-hip:
+	$(PYTHON) -c "from setup import build_hip; print(build_hip('gfx950'))"
+
+test:
+	$(PYTHON) -m pytest tests/ -q -m "not gpu"
+
+test-gpu:
+	$(PYTHON) -m pytest tests/ -q -m gpu
+
+bench:
+	$(PYTHON) bench.py --steps 30 --warmup 5
+
+image:
+	docker build -t $(IMAGE) .
+
+clean:
+	rm -rf build kata_xpu_device_plugin_amd/*.so kata_xpu_device_plugin_amd/**/__pycache__
