@@ -1,0 +1,87 @@
+"""CLI tool tests against the mock node: topo snapshot, bind, sriov, ident."""
+import json
+import os
+
+import pytest
+
+from kata_xpu_device_plugin_amd.config import Config
+from kata_xpu_device_plugin_amd.testing.mocknode import MockGPU, make_mock_node
+from kata_xpu_device_plugin_amd.tools.bind import Binder
+from kata_xpu_device_plugin_amd.tools.ident import collect
+from kata_xpu_device_plugin_amd.tools.sriov import set_numvfs, vf_bdfs
+from kata_xpu_device_plugin_amd.tools.topo import build_snapshot
+
+
+def test_topo_snapshot(tmp_path):
+    node = make_mock_node(str(tmp_path), n_gpus=8,
+                          hives=[[0, 1, 2, 3], [4, 5, 6, 7]], driver="amdgpu")
+    doc = build_snapshot(node.sysfs)
+    assert len(doc["gpus"]) == 8
+    assert len(doc["hives"]) == 2
+    assert all(len(h) == 4 for h in doc["hives"])
+    assert doc["xgmi_link_gbps"] == 153.0
+    assert doc["gpus"][0]["gfx_target_version"] == 90500
+
+
+def test_topo_snapshot_feeds_hint_loader(tmp_path):
+    from kata_xpu_device_plugin_amd.topology.hive import topology_from_hint
+    node = make_mock_node(str(tmp_path), n_gpus=4, driver="amdgpu")
+    doc = build_snapshot(node.sysfs)
+    hint = tmp_path / "hint.json"
+    hint.write_text(json.dumps(doc))
+    topo = topology_from_hint(str(hint))
+    assert topo is not None
+    assert len(topo.hive_of) == 4
+
+
+def test_ident_collect(tmp_path):
+    node = make_mock_node(str(tmp_path), n_gpus=2)
+    cfg = node.config()
+    doc = collect(cfg)
+    assert len(doc["functions"]) == 2
+    assert doc["functions"][0]["model_name"] == "INSTINCT_MI355X"
+
+
+def test_binder_status_and_rebind(tmp_path):
+    node = make_mock_node(str(tmp_path), n_gpus=2, driver="amdgpu",
+                          kfd=False, hint=False)
+    cfg = node.config()
+    b = Binder(cfg)
+    gpus = b.gpu_functions()
+    assert len(gpus) == 2
+    # mock sysfs has no writable unbind/probe files → dry-run only
+    b_dry = Binder(cfg, dry_run=True)
+    b_dry.rebind(gpus[0].bdf, "vfio-pci")  # must not raise
+    with pytest.raises(RuntimeError, match="no such device"):
+        b_dry.rebind("0000:ff:00.0", "vfio-pci")
+
+
+def test_sriov_enable_validation(tmp_path):
+    node = make_mock_node(str(tmp_path), n_gpus=1, driver="amdgpu",
+                          kfd=False, hint=False)
+    pf = node.gpus[0]
+    cfg = node.config()
+    with pytest.raises(RuntimeError, match="does not support SR-IOV"):
+        set_numvfs(cfg, pf.bdf, 4)
+    d = os.path.join(node.sysfs, "bus", "pci", "devices", pf.bdf)
+    open(os.path.join(d, "sriov_totalvfs"), "w").write("8\n")
+    open(os.path.join(d, "sriov_numvfs"), "w").write("0\n")
+    with pytest.raises(RuntimeError, match="at most 8"):
+        set_numvfs(cfg, pf.bdf, 16)
+    set_numvfs(cfg, pf.bdf, 4)
+    assert open(os.path.join(d, "sriov_numvfs")).read() == "4"
+
+
+def test_vf_bdfs_listing(tmp_path):
+    node = make_mock_node(str(tmp_path), n_gpus=1, driver="amdgpu",
+                          kfd=False, hint=False)
+    pf = node.gpus[0]
+    cfg = node.config()
+    d = os.path.join(node.sysfs, "bus", "pci", "devices", pf.bdf)
+    for k in range(2):
+        vf_bdf = f"0000:0a:02.{k}"
+        node.add_gpu(MockGPU(bdf=vf_bdf, device_id=0x75B3,
+                             iommu_group=str(100 + k), physfn_bdf=pf.bdf))
+        os.symlink(os.path.join(node.sysfs, "bus", "pci", "devices", vf_bdf),
+                   os.path.join(d, f"virtfn{k}"))
+    assert vf_bdfs(cfg, pf.bdf) == ["0000:0a:02.0", "0000:0a:02.1"]
