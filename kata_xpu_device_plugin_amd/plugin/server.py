@@ -47,6 +47,11 @@ from .state import DeviceState
 
 log = get_logger(__name__)
 
+try:
+    from .. import _native as _NATIVE
+except ImportError:  # pure-Python fallback (CPU test environments)
+    _NATIVE = None
+
 # Env var consumed by the Kata-side runtime to locate the CDI vendor class
 # (reference contract: generic_device_plugin.go:30-31, :348-350).
 ENV_CDI_VENDOR_CLASS = "KUBERNETES_CDI_VENDOR_CLASS"
@@ -84,6 +89,18 @@ class XPUDevicePlugin:
         self.allocations = 0          # metrics
         self.allocate_failures = 0
         self.last_allocate_s = 0.0
+        # hot-path precomputation
+        self._devices_dir = os.path.join(cfg.sysfs_root, "bus", "pci", "devices")
+        self._vendor_list = list(cfg.vendor_allowlist)
+        self._qn_cache: Dict[str, str] = {}      # gid → "kind=gid"
+        self._env_res_name = ENV_PCI_RESOURCE_PREFIX + resource_name.upper(
+        ).replace("/", "_").replace(".", "_").replace("-", "_")
+
+    def _qn(self, gid: str) -> str:
+        qn = self._qn_cache.get(gid)
+        if qn is None:
+            qn = self._qn_cache[gid] = qualified_name(self.cfg.cdi_kind, gid)
+        return qn
 
     # ------------------------------------------------------------------
     # DevicePlugin service
@@ -128,29 +145,35 @@ class XPUDevicePlugin:
             self.state.unwatch(q)
 
     # -- Allocate ------------------------------------------------------
-    def _revalidate(self, gid: str) -> None:
-        """Check the group still exists on this node with our vendor bound
-        to vfio (reference re-reads iommu_group+vendor per function:
-        generic_device_plugin.go:329-338). Native C++ path when built."""
-        dev = self.state.device(gid)
-        if dev is None:
-            raise AllocationError(f"unknown device id {gid}")
-        devices_dir = os.path.join(self.cfg.sysfs_root, "bus", "pci", "devices")
+    def _revalidate_many(self, ids: List[str]) -> None:
+        """Check every requested group still exists on this node with our
+        vendor bound to vfio (reference re-reads iommu_group+vendor per
+        function: generic_device_plugin.go:329-338). One batched native
+        call per container request when the C++ extension is built."""
+        groups = []
+        for gid in ids:
+            dev = self.state.device(gid)
+            if dev is None:
+                raise AllocationError(f"unknown device id {gid}")
+            groups.append((gid, [fn.bdf for fn in dev.functions]))
         if self.cfg.native != "off":
-            try:
-                from .. import _native
-            except ImportError:
-                _native = None
-                if self.cfg.native == "require":
-                    raise RuntimeError("_native extension required but missing")
-            if _native is not None:
-                err = _native.revalidate_group(
-                    devices_dir, gid, [fn.bdf for fn in dev.functions],
-                    list(self.cfg.vendor_allowlist), self.cfg.required_driver,
+            if _NATIVE is None and self.cfg.native == "require":
+                raise RuntimeError("_native extension required but missing")
+            if _NATIVE is not None:
+                err = _NATIVE.revalidate_groups(
+                    self._devices_dir, groups,
+                    self._vendor_list, self.cfg.required_driver,
                 )
                 if err:
                     raise AllocationError(err)
                 return
+        for gid, _bdfs in groups:
+            self._revalidate_py(gid)
+
+    def _revalidate_py(self, gid: str) -> None:
+        dev = self.state.device(gid)
+        assert dev is not None
+        devices_dir = self._devices_dir
         for fn in dev.functions:
             p = os.path.join(devices_dir, fn.bdf)
             if read_link_base(os.path.join(p, "iommu_group")) != gid:
@@ -172,10 +195,10 @@ class XPUDevicePlugin:
             assert dev is not None  # validated earlier
             bdfs.extend(dev.bdfs)
             if strategy == STRATEGY_CDI_CRI:
-                resp.cdi_devices.add(name=qualified_name(self.cfg.cdi_kind, gid))
+                resp.cdi_devices.add(name=self._qn(gid))
             elif strategy == STRATEGY_CDI_ANNOTATIONS:
                 key = f"{ANNOTATION_PREFIX}vfio{gid}"
-                resp.annotations[key] = qualified_name(self.cfg.cdi_kind, gid)
+                resp.annotations[key] = self._qn(gid)
             else:  # raw device nodes, no CDI
                 path = os.path.join(self.cfg.dev_root, dev.vfio_node)
                 resp.devices.add(
@@ -183,9 +206,7 @@ class XPUDevicePlugin:
                 )
         if strategy in (STRATEGY_CDI_CRI, STRATEGY_CDI_ANNOTATIONS):
             resp.envs[ENV_CDI_VENDOR_CLASS] = self.cfg.cdi_kind
-        env_res = ENV_PCI_RESOURCE_PREFIX + self.resource_name.upper().replace(
-            "/", "_").replace(".", "_").replace("-", "_")
-        resp.envs[env_res] = ",".join(bdfs)
+        resp.envs[self._env_res_name] = ",".join(bdfs)
         return resp
 
     def Allocate(self, request, context):
@@ -196,8 +217,7 @@ class XPUDevicePlugin:
         try:
             for creq in request.container_requests:
                 ids = list(creq.devices_ids)
-                for gid in ids:
-                    self._revalidate(gid)
+                self._revalidate_many(ids)
                 response.container_responses.append(self._container_response(ids))
         except AllocationError as e:
             self.allocate_failures += 1

@@ -187,6 +187,46 @@ std::string revalidate_group(const std::string& devices_dir,
     return err;
 }
 
+// Batched form: validate several groups with ONE directory fd — the
+// Allocate() hot path calls this once per container request.
+std::string revalidate_groups(
+    const std::string& devices_dir,
+    const std::vector<std::pair<std::string, std::vector<std::string>>>& groups,
+    const std::vector<long>& vendors,
+    const std::string& required_driver) {
+    int base_fd = ::open(devices_dir.c_str(), O_RDONLY | O_DIRECTORY | O_CLOEXEC);
+    if (base_fd < 0) return "cannot open " + devices_dir;
+    std::string err;
+    for (const auto& [gid, bdfs] : groups) {
+        for (const auto& bdf : bdfs) {
+            int dfd = ::openat(base_fd, bdf.c_str(),
+                               O_RDONLY | O_DIRECTORY | O_CLOEXEC);
+            if (dfd < 0) { err = "device " + bdf + " vanished"; goto done; }
+            if (link_base(dfd, "iommu_group") != gid) {
+                ::close(dfd);
+                err = "device " + bdf + " no longer in IOMMU group " + gid;
+                goto done;
+            }
+            auto vendor = parse_hex(read_attr(dfd, "vendor"));
+            if (!vendor || std::find(vendors.begin(), vendors.end(), *vendor) ==
+                               vendors.end()) {
+                ::close(dfd);
+                err = "device " + bdf + " vendor not allowed";
+                goto done;
+            }
+            if (link_base(dfd, "driver") != required_driver) {
+                ::close(dfd);
+                err = "device " + bdf + " not bound to " + required_driver;
+                goto done;
+            }
+            ::close(dfd);
+        }
+    }
+done:
+    ::close(base_fd);
+    return err;
+}
+
 // ---------------------------------------------------------------------------
 // Preferred-set selection: exact composition search over (hive, numa)
 // buckets; must mirror topology/hive.py::preferred_sets (tested for parity).
@@ -344,6 +384,9 @@ PYBIND11_MODULE(_native, m) {
           py::arg("group_id"), py::arg("bdfs"), py::arg("vendors"),
           py::arg("required_driver"),
           "Allocate-path group revalidation; empty string = OK");
+    m.def("revalidate_groups", &revalidate_groups, py::arg("devices_dir"),
+          py::arg("groups"), py::arg("vendors"), py::arg("required_driver"),
+          "Batched group revalidation (one dirfd); empty string = OK");
     m.def("select_preferred", &select_preferred, py::arg("locality"),
           py::arg("available"), py::arg("must_include"), py::arg("size"),
           "Exact xGMI/NUMA bucket-composition preferred-set selection");

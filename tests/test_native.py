@@ -109,3 +109,16 @@ def test_selector_parity_mock_node(tmp_path):
         nat = preferred_sets(topo, bdf_of, ids, [], k, use_native=True)
         assert score_set(topo, [bdf_of[d] for d in py]) == \
             score_set(topo, [bdf_of[d] for d in nat])
+
+
+def test_revalidate_groups_batched(tmp_path):
+    node = make_mock_node(str(tmp_path), n_gpus=4, kfd=False, hint=False)
+    cfg = node.config()
+    devices_dir = f"{cfg.sysfs_root}/bus/pci/devices"
+    groups = [(str(70 + i), [f"0000:{0x0a + 8 * i:02x}:00.0"]) for i in range(4)]
+    assert _native.revalidate_groups(devices_dir, groups, [0x1002], "vfio-pci") == ""
+    bad = groups + [("99", ["0000:0a:00.0"])]
+    assert "no longer in IOMMU group" in _native.revalidate_groups(
+        devices_dir, bad, [0x1002], "vfio-pci")
+    assert "vanished" in _native.revalidate_groups(
+        devices_dir, [("70", ["0000:ff:00.0"])], [0x1002], "vfio-pci")
