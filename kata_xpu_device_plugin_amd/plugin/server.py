@@ -86,6 +86,9 @@ class XPUDevicePlugin:
         self._server: Optional[grpc.Server] = None
         self._stop = threading.Event()
         self._serving = threading.Event()
+        # start/stop/restart may be invoked from the manager thread AND the
+        # health watcher (socket-removal recovery) concurrently; serialize.
+        self._lifecycle = threading.Lock()
         self.allocations = 0          # metrics
         self.allocate_failures = 0
         self.last_allocate_s = 0.0
@@ -251,6 +254,10 @@ class XPUDevicePlugin:
     # generic_device_plugin.go:128-219)
     # ------------------------------------------------------------------
     def start(self, register: bool = True) -> None:
+        with self._lifecycle:
+            self._start_locked(register)
+
+    def _start_locked(self, register: bool) -> None:
         self._stop.clear()
         server = grpc.server(
             futures.ThreadPoolExecutor(max_workers=8),
@@ -295,6 +302,10 @@ class XPUDevicePlugin:
             ch.close()
 
     def stop(self) -> None:
+        with self._lifecycle:
+            self._stop_locked()
+
+    def _stop_locked(self) -> None:
         self._stop.set()
         self._serving.clear()
         if self._server is not None:

@@ -1,0 +1,187 @@
+"""Robustness / end-to-end edge cases: multi-model nodes, concurrent
+health flapping under churn, rescans, multi-container Allocate."""
+import os
+import threading
+import time
+
+import grpc
+import pytest
+
+from kata_xpu_device_plugin_amd.cdi.spec import read_spec
+from kata_xpu_device_plugin_amd.plugin import api
+from kata_xpu_device_plugin_amd.plugin.manager import PluginManager
+from kata_xpu_device_plugin_amd.testing.kubelet_stub import KubeletStub
+from kata_xpu_device_plugin_amd.testing.mocknode import MockGPU, make_mock_node
+
+
+@pytest.fixture
+def multi_model(tmp_path):
+    """Node with 6 MI355X + 2 MI300X (two resource names → two plugins,
+    like the reference's per-model plugin split, device_plugin.go:83-119)."""
+    node = make_mock_node(str(tmp_path), n_gpus=6, kfd=False, hint=False)
+    for k in range(2):
+        node.add_gpu(MockGPU(bdf=f"0000:e{k}:00.0", device_id=0x74A1,
+                             iommu_group=str(90 + k), numa_node=1))
+    cfg = node.config()
+    stub = KubeletStub(cfg.kubelet_socket_dir)
+    stub.start()
+    mgr = PluginManager(cfg)
+    mgr.setup()
+    mgr.start(register=True)
+    yield node, cfg, stub, mgr
+    mgr.stop()
+    stub.stop()
+
+
+def test_two_models_two_plugins(multi_model):
+    node, cfg, stub, mgr = multi_model
+    regs = stub.wait_for_registration(2)
+    names = sorted(r.resource_name for r in regs)
+    assert names == ["amd.com/INSTINCT_MI300X", "amd.com/INSTINCT_MI355X"]
+    # distinct sockets
+    assert len({r.endpoint for r in regs}) == 2
+    by_name = {r.resource_name: r for r in regs}
+    ps300 = stub.plugin_stub(by_name["amd.com/INSTINCT_MI300X"].endpoint)
+    stream = ps300.ListAndWatch(api.Empty())
+    devs = next(stream).devices
+    assert [d.id for d in devs] == ["90", "91"]
+    stream.cancel()
+    # cross-model allocate must fail (id belongs to the other plugin)
+    with pytest.raises(grpc.RpcError):
+        ps300.Allocate(api.AllocateRequest(container_requests=[
+            api.ContainerAllocateRequest(devices_ids=["70"])]))
+
+
+def test_multi_container_request(multi_model):
+    node, cfg, stub, mgr = multi_model
+    regs = stub.wait_for_registration(2)
+    r = next(x for x in regs if x.resource_name.endswith("MI355X"))
+    ps = stub.plugin_stub(r.endpoint)
+    resp = ps.Allocate(api.AllocateRequest(container_requests=[
+        api.ContainerAllocateRequest(devices_ids=["70", "71"]),
+        api.ContainerAllocateRequest(devices_ids=["72"]),
+    ]))
+    assert len(resp.container_responses) == 2
+    assert len(resp.container_responses[0].cdi_devices) == 2
+    assert len(resp.container_responses[1].cdi_devices) == 1
+
+
+def test_cdi_spec_covers_all_models(multi_model):
+    node, cfg, stub, mgr = multi_model
+    spec = read_spec(mgr.cdi_spec_path)
+    assert sorted(spec.device_names(), key=int) == [
+        "70", "71", "72", "73", "74", "75", "90", "91"]
+
+
+def test_rescan_rewrites_cdi(multi_model):
+    node, cfg, stub, mgr = multi_model
+    node.add_gpu(MockGPU(bdf="0000:f5:00.0", iommu_group="95"))
+    mgr.rescan()
+    spec = read_spec(mgr.cdi_spec_path)
+    assert "95" in spec.device_names()
+    # served immediately on the existing plugin
+    st = mgr.states["amd.com/INSTINCT_MI355X"]
+    assert "95" in st.device_ids()
+
+
+def test_health_flap_during_churn(tmp_path):
+    """Concurrent Allocate churn + vfio node flapping: no crashes, flapped
+    device rejected only while its sysfs state is intact (health is a
+    ListAndWatch concern; Allocate revalidates sysfs, not /dev)."""
+    node = make_mock_node(str(tmp_path), n_gpus=4)
+    cfg = node.config()
+    stub = KubeletStub(cfg.kubelet_socket_dir)
+    stub.start()
+    mgr = PluginManager(cfg)
+    mgr.setup()
+    mgr.start()
+    try:
+        r = stub.wait_for_registration(1)[0]
+        ps = stub.plugin_stub(r.endpoint)
+        stream = ps.ListAndWatch(api.Empty())
+        next(stream)
+        stop = threading.Event()
+        errs = []
+
+        def churn():
+            try:
+                while not stop.is_set():
+                    ps.Allocate(api.AllocateRequest(container_requests=[
+                        api.ContainerAllocateRequest(devices_ids=["70"])]))
+            except Exception as e:  # pragma: no cover
+                errs.append(e)
+
+        t = threading.Thread(target=churn)
+        t.start()
+        seen_unhealthy = seen_healthy = False
+        for _ in range(6):
+            node.remove_vfio_node("72")
+            time.sleep(0.05)
+            node.add_vfio_node("72")
+            time.sleep(0.05)
+        deadline = time.monotonic() + 5
+        while time.monotonic() < deadline and not (seen_unhealthy and seen_healthy):
+            upd = next(stream)
+            h = {d.id: d.health for d in upd.devices}
+            if h.get("72") == api.UNHEALTHY:
+                seen_unhealthy = True
+            if seen_unhealthy and h.get("72") == api.HEALTHY:
+                seen_healthy = True
+        stop.set()
+        t.join()
+        stream.cancel()
+        assert not errs
+        assert seen_unhealthy and seen_healthy
+    finally:
+        mgr.stop()
+        stub.stop()
+
+
+def test_plugin_restart_keeps_serving(tmp_path):
+    node = make_mock_node(str(tmp_path), n_gpus=2)
+    cfg = node.config()
+    stub = KubeletStub(cfg.kubelet_socket_dir)
+    stub.start()
+    mgr = PluginManager(cfg)
+    mgr.setup()
+    mgr.start()
+    try:
+        r = stub.wait_for_registration(1)[0]
+        plugin = next(iter(mgr.plugins.values()))
+        for _ in range(3):
+            plugin.restart()
+        regs = stub.wait_for_registration(4)  # initial + 3 restarts
+        assert len(regs) >= 4
+        # fresh channel post-restart
+        ch = grpc.insecure_channel(f"unix://{plugin.socket_path}")
+        grpc.channel_ready_future(ch).result(timeout=5)
+        ps = api.DevicePluginStub(ch)
+        resp = ps.Allocate(api.AllocateRequest(container_requests=[
+            api.ContainerAllocateRequest(devices_ids=["70"])]))
+        assert resp.container_responses[0].cdi_devices[0].name == "amd.com/gpu=70"
+        ch.close()
+    finally:
+        mgr.stop()
+        stub.stop()
+
+
+def test_allocate_empty_request(tmp_path):
+    node = make_mock_node(str(tmp_path), n_gpus=1)
+    cfg = node.config()
+    stub = KubeletStub(cfg.kubelet_socket_dir)
+    stub.start()
+    mgr = PluginManager(cfg)
+    mgr.setup()
+    mgr.start()
+    try:
+        r = stub.wait_for_registration(1)[0]
+        ps = stub.plugin_stub(r.endpoint)
+        resp = ps.Allocate(api.AllocateRequest())
+        assert len(resp.container_responses) == 0
+        resp = ps.Allocate(api.AllocateRequest(container_requests=[
+            api.ContainerAllocateRequest()]))
+        assert len(resp.container_responses) == 1
+        assert len(resp.container_responses[0].cdi_devices) == 0
+    finally:
+        mgr.stop()
+        stub.stop()
