@@ -43,6 +43,8 @@ class NodeWatcher(threading.Thread):
         on_socket_removed: Optional[Callable[[str], None]] = None,   # socket name
         on_kubelet_restarted: Optional[Callable[[], None]] = None,
         plugin_socket_names: Optional[set] = None,
+        on_cdi_spec_removed: Optional[Callable[[str], None]] = None,  # file name
+        cdi_spec_names: Optional[set] = None,
     ):
         super().__init__(name="kxdp-node-watcher", daemon=True)
         self.cfg = cfg
@@ -50,6 +52,8 @@ class NodeWatcher(threading.Thread):
         self.on_socket_removed = on_socket_removed
         self.on_kubelet_restarted = on_kubelet_restarted
         self.plugin_socket_names = plugin_socket_names or set()
+        self.on_cdi_spec_removed = on_cdi_spec_removed
+        self.cdi_spec_names = cdi_spec_names or set()
         self._stop_evt = threading.Event()
         self._ready = threading.Event()
 
@@ -65,18 +69,21 @@ class NodeWatcher(threading.Thread):
         os.makedirs(vfio_dir, exist_ok=True)
         with inotify.Inotify() as ino:
             wd_vfio = ino.add_watch(vfio_dir, _MASK_DIR)
-            wd_kubelet = -1
+            wd_kubelet = wd_cdi = -1
             if os.path.isdir(self.cfg.kubelet_socket_dir):
                 wd_kubelet = ino.add_watch(self.cfg.kubelet_socket_dir, _MASK_DIR)
+            if os.path.isdir(self.cfg.cdi_dir):
+                wd_cdi = ino.add_watch(self.cfg.cdi_dir, _MASK_DIR)
             self._ready.set()
             while not self._stop_evt.is_set():
                 for ev in ino.read_events(timeout=0.2):
                     try:
-                        self._handle(ev, wd_vfio, wd_kubelet)
+                        self._handle(ev, wd_vfio, wd_kubelet, wd_cdi)
                     except Exception:  # watcher must never die silently
                         log.exception("health watcher event error: %s", ev)
 
-    def _handle(self, ev: inotify.Event, wd_vfio: int, wd_kubelet: int) -> None:
+    def _handle(self, ev: inotify.Event, wd_vfio: int, wd_kubelet: int,
+                wd_cdi: int = -1) -> None:
         if ev.wd == wd_vfio:
             st = self._owner(ev.name)
             if st is None:
@@ -94,6 +101,14 @@ class NodeWatcher(threading.Thread):
                 log.warning("kubelet.sock re-created; kubelet restarted")
                 if self.on_kubelet_restarted:
                     self.on_kubelet_restarted()
+        elif ev.wd == wd_cdi:
+            # self-heal: the CDI spec is the runtime's source of truth for
+            # resolving our device names; if an operator or tmp-cleaner
+            # removes it, regenerate immediately.
+            if ev.name in self.cdi_spec_names and ev.removed:
+                log.warning("CDI spec %s removed; regenerating", ev.name)
+                if self.on_cdi_spec_removed:
+                    self.on_cdi_spec_removed(ev.name)
 
     def wait_ready(self, timeout: float = 5.0) -> None:
         if not self._ready.wait(timeout):

@@ -8,6 +8,7 @@ a single health watcher, and re-registration on kubelet restart.
 """
 from __future__ import annotations
 
+import os
 import signal
 import threading
 from typing import Dict, List, Optional
@@ -93,6 +94,9 @@ class PluginManager:
             on_socket_removed=self._on_socket_removed,
             on_kubelet_restarted=self._on_kubelet_restarted,
             plugin_socket_names={p.socket_name for p in self.plugins.values()},
+            on_cdi_spec_removed=self._on_cdi_spec_removed,
+            cdi_spec_names={os.path.basename(self.cdi_spec_path)}
+            if self.cdi_spec_path else set(),
         )
         self.watcher.start()
         self.watcher.wait_ready()
@@ -114,6 +118,17 @@ class PluginManager:
                     plugin.restart()
                 except Exception:
                     log.exception("restart of %s failed", plugin.resource_name)
+
+    def _on_cdi_spec_removed(self, name: str) -> None:
+        if self._stop.is_set() or self.inventory is None:
+            return
+        try:
+            spec = build_spec(self.inventory, self.cfg.cdi_kind, self.cfg.dev_root)
+            self.cdi_spec_path = write_spec(
+                spec, self.cfg.cdi_dir, self.cfg.cdi_spec_name, self.cfg.cdi_format
+            )
+        except Exception:
+            log.exception("CDI spec regeneration failed")
 
     def _on_amdsmi_health(self, bdf: str, healthy: bool, reasons) -> None:
         """amd-smi verdict → DeviceState (matches any function BDF of a

@@ -185,3 +185,25 @@ def test_allocate_empty_request(tmp_path):
     finally:
         mgr.stop()
         stub.stop()
+
+
+def test_cdi_spec_self_heal(tmp_path):
+    """Deleting the CDI spec file regenerates it (runtime resolution must
+    never dangle)."""
+    node = make_mock_node(str(tmp_path), n_gpus=2)
+    cfg = node.config()
+    mgr = PluginManager(cfg)
+    mgr.setup()
+    mgr.start(register=False)
+    try:
+        path = mgr.cdi_spec_path
+        assert os.path.exists(path)
+        os.unlink(path)
+        deadline = time.monotonic() + 5
+        while not os.path.exists(path) and time.monotonic() < deadline:
+            time.sleep(0.05)
+        assert os.path.exists(path), "CDI spec must be regenerated"
+        spec = read_spec(path)
+        assert spec.device_names() == ["70", "71"]
+    finally:
+        mgr.stop()
