@@ -115,6 +115,10 @@ class Config:
     # --- health ---
     health_poll_interval_s: float = field(default_factory=lambda: _env_float("HEALTH_POLL_S", 10.0))
     amdsmi_health: bool = field(default_factory=lambda: _env_bool("AMDSMI_HEALTH", True))
+    # Reject Allocate() for devices currently marked Unhealthy (kubelet
+    # should not send them, but a stale scheduler view can; defensive
+    # default on). The reference allocates regardless of health.
+    reject_unhealthy: bool = field(default_factory=lambda: _env_bool("REJECT_UNHEALTHY", True))
 
     # --- observability ---
     metrics_port: int = field(default_factory=lambda: _env_int("METRICS_PORT", 0))  # 0 = off

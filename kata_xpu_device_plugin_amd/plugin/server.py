@@ -158,6 +158,8 @@ class XPUDevicePlugin:
             dev = self.state.device(gid)
             if dev is None:
                 raise AllocationError(f"unknown device id {gid}")
+            if self.cfg.reject_unhealthy and not self.state.is_healthy(gid):
+                raise AllocationError(f"device {gid} is Unhealthy")
             groups.append((gid, [fn.bdf for fn in dev.functions]))
         if self.cfg.native != "off":
             if _NATIVE is None and self.cfg.native == "require":
