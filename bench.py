@@ -160,6 +160,17 @@ def main() -> int:
             if use_cuda:
                 torch.cuda.synchronize()
 
+    # RPC floor: a no-op unary RPC on the same channel (contextualizes how
+    # much of the Allocate latency is grpc transport vs. plugin work).
+    for _ in range(20):
+        plugin.GetDevicePluginOptions(api.Empty())
+    floor = []
+    for _ in range(100):
+        f0 = time.perf_counter()
+        plugin.GetDevicePluginOptions(api.Empty())
+        floor.append(time.perf_counter() - f0)
+    rpc_floor_us = statistics.median(floor) * 1e6
+
     # warmup (untimed)
     run_steps(args.warmup)
 
@@ -197,6 +208,7 @@ def main() -> int:
             "pods_per_step": args.pods_per_step,
             "concurrent_clients": world_size,
             "admission_p50_us": round(adm_p50_us, 1),  # preferred+allocate
+            "rpc_floor_us": round(rpc_floor_us, 1),    # no-op RPC, same channel
             "p99_us": round(p99_us, 1),
             "pods_per_s_total": round(pods_per_s, 1),
             "parallelism": f"{world_size} churn client(s), one plugin daemon",
