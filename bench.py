@@ -160,19 +160,18 @@ def main() -> int:
             if use_cuda:
                 torch.cuda.synchronize()
 
-    # RPC floor: a no-op unary RPC on the same channel (contextualizes how
-    # much of the Allocate latency is grpc transport vs. plugin work).
-    for _ in range(20):
-        plugin.GetDevicePluginOptions(api.Empty())
+    # warmup (untimed)
+    run_steps(args.warmup)
+
+    # RPC floor: a no-op unary RPC on the same (now warm) channel —
+    # contextualizes how much of the Allocate latency is grpc transport
+    # vs. plugin work.
     floor = []
-    for _ in range(100):
+    for _ in range(200):
         f0 = time.perf_counter()
         plugin.GetDevicePluginOptions(api.Empty())
         floor.append(time.perf_counter() - f0)
     rpc_floor_us = statistics.median(floor) * 1e6
-
-    # warmup (untimed)
-    run_steps(args.warmup)
 
     barrier_sync()
     t_start = time.perf_counter()
