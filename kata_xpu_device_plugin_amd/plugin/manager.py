@@ -140,7 +140,7 @@ class PluginManager:
                 if dev and bdf in (fn.bdf.lower() for fn in dev.functions):
                     if not healthy:
                         log.warning("amd-smi: %s unhealthy: %s", bdf, reasons)
-                    state.set_health(gid, healthy)
+                    state.set_health(gid, healthy, source="amdsmi")
                     return
 
     def _on_kubelet_restarted(self) -> None:

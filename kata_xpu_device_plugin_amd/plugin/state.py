@@ -24,15 +24,19 @@ class DeviceState:
     def __init__(self, devices: Dict[str, XPUDevice]):
         self._lock = threading.Lock()
         self._devices: Dict[str, XPUDevice] = dict(devices)
-        self._healthy: Dict[str, bool] = {gid: True for gid in devices}
+        # Health is tracked per SOURCE ("vfio" watcher, "amdsmi" poller,
+        # "probe" burn-in, ...); a device is Healthy iff no source holds it
+        # unhealthy — so an amd-smi recovery cannot mask a missing
+        # /dev/vfio node and vice versa.
+        self._unhealthy_by: Dict[str, set] = {gid: set() for gid in devices}
         self._watchers: List[queue.Queue] = []
         self._generation = 0
 
     # -- snapshots ------------------------------------------------------
     def snapshot(self) -> List[Tuple[XPUDevice, bool]]:
         with self._lock:
-            return [(self._devices[g], self._healthy[g]) for g in sorted(
-                self._devices, key=_gkey)]
+            return [(self._devices[g], not self._unhealthy_by[g])
+                    for g in sorted(self._devices, key=_gkey)]
 
     def device(self, gid: str) -> Optional[XPUDevice]:
         with self._lock:
@@ -44,30 +48,42 @@ class DeviceState:
 
     def healthy_ids(self) -> List[str]:
         with self._lock:
-            return sorted((g for g, h in self._healthy.items() if h), key=_gkey)
+            return sorted((g for g, u in self._unhealthy_by.items() if not u),
+                          key=_gkey)
 
     def is_healthy(self, gid: str) -> bool:
         with self._lock:
-            return self._healthy.get(gid, False)
+            u = self._unhealthy_by.get(gid)
+            return u is not None and not u
 
     # -- mutation -------------------------------------------------------
-    def set_health(self, gid: str, healthy: bool) -> bool:
-        """Returns True if the state changed (and watchers were notified)."""
+    def set_health(self, gid: str, healthy: bool, source: str = "vfio") -> bool:
+        """Record one source's verdict; returns True if the device's
+        EFFECTIVE health changed (and watchers were notified)."""
         with self._lock:
-            if gid not in self._devices or self._healthy.get(gid) == healthy:
+            if gid not in self._devices:
                 return False
-            self._healthy[gid] = healthy
+            u = self._unhealthy_by[gid]
+            before = not u
+            if healthy:
+                u.discard(source)
+            else:
+                u.add(source)
+            after = not u
+            if before == after:
+                return False
             self._generation += 1
             self._notify_locked()
-        log.info("device %s → %s", gid, "Healthy" if healthy else "Unhealthy")
+        log.info("device %s → %s (source %s)", gid,
+                 "Healthy" if after else "Unhealthy", source)
         return True
 
     def replace_devices(self, devices: Dict[str, XPUDevice]) -> None:
         """Swap in a fresh discovery result (rescan)."""
         with self._lock:
-            old_health = self._healthy
+            old = self._unhealthy_by
             self._devices = dict(devices)
-            self._healthy = {g: old_health.get(g, True) for g in devices}
+            self._unhealthy_by = {g: set(old.get(g, set())) for g in devices}
             self._generation += 1
             self._notify_locked()
 

@@ -261,3 +261,21 @@ def test_metrics_exporter_refresh(tmp_path):
     text = generate_latest(exp.registry).decode()
     assert 'kxdp_devices{resource="amd.com/INSTINCT_MI355X"} 2.0' in text
     assert "kxdp_discovery_seconds" in text
+
+
+def test_state_multi_source_health(tmp_path):
+    """A device is Healthy iff NO source holds it unhealthy; one source's
+    recovery cannot mask another's outstanding failure."""
+    st, inv = _state_of(tmp_path)
+    assert st.set_health("70", False, source="vfio")
+    assert not st.is_healthy("70")
+    # amd-smi reports healthy — vfio verdict still stands
+    assert not st.set_health("70", True, source="amdsmi")
+    assert not st.is_healthy("70")
+    # amd-smi also fails, then vfio recovers: still unhealthy via amdsmi
+    st.set_health("70", False, source="amdsmi")
+    assert not st.set_health("70", True, source="vfio")
+    assert not st.is_healthy("70")
+    # last source clears → healthy
+    assert st.set_health("70", True, source="amdsmi")
+    assert st.is_healthy("70")
