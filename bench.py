@@ -58,10 +58,14 @@ def main() -> int:
         torch.cuda.set_device(local_rank % max(1, torch.cuda.device_count()))
 
     if distributed:
+        # RCCL needs one GPU per rank; fall back to gloo when testing more
+        # clients than GPUs (e.g. a 1-GPU box).
+        nccl_ok = use_cuda and torch.cuda.device_count() >= world_size
         dist.init_process_group(
-            backend="nccl" if use_cuda else "gloo",
+            backend="nccl" if nccl_ok else "gloo",
             rank=rank, world_size=world_size,
         )
+        use_cuda = nccl_ok
 
     from kata_xpu_device_plugin_amd.plugin import api
     from kata_xpu_device_plugin_amd.plugin.manager import PluginManager
