@@ -102,6 +102,14 @@ class Config:
     )
     kubelet_socket_name: str = "kubelet.sock"
     plugin_socket_prefix: str = field(default_factory=lambda: _env("SOCKET_PREFIX", "kata-xpu-amd"))
+    # How kubelet learns about us: "legacy" self-registration against
+    # kubelet.sock (the reference's only mode), "watcher" = serve the
+    # pluginregistration.v1 service under plugins_registry/ for kubelet's
+    # plugin watcher, or "both".
+    registration_mode: str = field(default_factory=lambda: _env("REGISTRATION_MODE", "legacy"))
+    plugins_registry_dir: str = field(
+        default_factory=lambda: _env("PLUGINS_REGISTRY", "/var/lib/kubelet/plugins_registry")
+    )
     grpc_timeout_s: float = field(default_factory=lambda: _env_float("GRPC_TIMEOUT_S", 5.0))
 
     # --- topology ---
@@ -141,6 +149,10 @@ class Config:
             raise ValueError(f"cdi_kind must look like vendor/class, got {self.cdi_kind!r}")
         if self.native not in ("auto", "require", "off"):
             raise ValueError(f"native must be auto|require|off, got {self.native!r}")
+        if self.registration_mode not in ("legacy", "watcher", "both"):
+            raise ValueError(
+                f"registration_mode must be legacy|watcher|both, "
+                f"got {self.registration_mode!r}")
 
     @property
     def kubelet_socket(self) -> str:

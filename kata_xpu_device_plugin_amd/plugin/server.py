@@ -92,6 +92,8 @@ class XPUDevicePlugin:
         # start/stop/restart may be invoked from the manager thread AND the
         # health watcher (socket-removal recovery) concurrently; serialize.
         self._lifecycle = threading.Lock()
+        self.watcher_servicer = None  # set in watcher registration mode
+        self.watcher_socket_path: Optional[str] = None
         self.allocations = 0          # metrics
         self.allocate_failures = 0
         self.last_allocate_s = 0.0
@@ -296,9 +298,11 @@ class XPUDevicePlugin:
         finally:
             ch.close()
         self._serving.set()
-        if register:
+        if register and self.cfg.registration_mode in ("legacy", "both"):
             self.register_with_kubelet()
-        log.info("plugin %s serving on %s", self.resource_name, self.socket_path)
+        log.info("plugin %s serving on %s (%s registration)",
+                 self.resource_name, self.socket_path,
+                 self.cfg.registration_mode)
 
     def _serve_thread(self) -> None:
         loop = asyncio.new_event_loop()
@@ -312,6 +316,20 @@ class XPUDevicePlugin:
             if os.path.exists(self.socket_path):
                 os.unlink(self.socket_path)
             server.add_insecure_port(f"unix://{self.socket_path}")
+            if self.cfg.registration_mode in ("watcher", "both"):
+                from .watcher_registration import (
+                    WatcherRegistrationServicer,
+                    add_watcher_registration_servicer,
+                )
+                os.makedirs(self.cfg.plugins_registry_dir, exist_ok=True)
+                self.watcher_servicer = WatcherRegistrationServicer(
+                    self.resource_name, self.socket_path)
+                self.watcher_socket_path = os.path.join(
+                    self.cfg.plugins_registry_dir, self.socket_name)
+                add_watcher_registration_servicer(server, self.watcher_servicer)
+                if os.path.exists(self.watcher_socket_path):
+                    os.unlink(self.watcher_socket_path)
+                server.add_insecure_port(f"unix://{self.watcher_socket_path}")
             await server.start()
             self._aio_server = server
             self._stop_async = asyncio.Event()
@@ -372,11 +390,13 @@ class XPUDevicePlugin:
             self._thread = None
         self._aio_server = None
         self._loop = None
-        if os.path.exists(self.socket_path):
-            try:
-                os.unlink(self.socket_path)
-            except OSError:
-                pass
+        for path in (self.socket_path,
+                     getattr(self, "watcher_socket_path", None)):
+            if path and os.path.exists(path):
+                try:
+                    os.unlink(path)
+                except OSError:
+                    pass
 
     def restart(self, register: bool = True) -> None:
         """Full stop/start cycle, e.g. after kubelet restart (reference

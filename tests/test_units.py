@@ -279,3 +279,53 @@ def test_state_multi_source_health(tmp_path):
     # last source clears → healthy
     assert st.set_health("70", True, source="amdsmi")
     assert st.is_healthy("70")
+
+
+# --- plugin-watcher registration ------------------------------------------
+
+def test_watcher_registration_mode(tmp_path):
+    """pluginregistration.v1: kubelet-side GetInfo/NotifyRegistrationStatus
+    against the plugin's registry socket (the reference supports only
+    legacy self-registration)."""
+    import grpc as _grpc
+    import tempfile as _tf
+    from kata_xpu_device_plugin_amd.plugin.watcher_registration import (
+        InfoRequest, RegistrationStatus, WatcherRegistrationStub)
+
+    node = make_mock_node(str(tmp_path), n_gpus=1)
+    registry = _tf.mkdtemp(prefix="kxdp-reg-")
+    cfg = node.config(registration_mode="watcher", plugins_registry_dir=registry)
+    mgr = PluginManager(cfg)
+    mgr.setup()
+    mgr.start(register=True)  # no kubelet.sock needed in watcher mode
+    try:
+        plugin = next(iter(mgr.plugins.values()))
+        assert plugin.watcher_socket_path and os.path.exists(plugin.watcher_socket_path)
+        ch = _grpc.insecure_channel(f"unix://{plugin.watcher_socket_path}")
+        _grpc.channel_ready_future(ch).result(timeout=5)
+        stub = WatcherRegistrationStub(ch)
+        info = stub.GetInfo(InfoRequest())
+        assert info.type == "DevicePlugin"
+        assert info.name == "amd.com/INSTINCT_MI355X"
+        assert info.endpoint == plugin.socket_path
+        assert list(info.supported_versions) == ["v1beta1"]
+        stub.NotifyRegistrationStatus(RegistrationStatus(plugin_registered=True))
+        assert plugin.watcher_servicer.last_status == (True, "")
+        # the DevicePlugin service itself is reachable on BOTH sockets
+        ps = api.DevicePluginStub(ch)
+        opts = ps.GetDevicePluginOptions(api.Empty())
+        assert opts.get_preferred_allocation_available
+        ch.close()
+    finally:
+        mgr.stop()
+
+
+def test_watcher_wire_format():
+    from kata_xpu_device_plugin_amd.plugin.watcher_registration import PluginInfo
+    msg = PluginInfo(type="DevicePlugin", name="amd.com/X", endpoint="/e.sock",
+                     supported_versions=["v1beta1"])
+    b = msg.SerializeToString()
+    def _ld(n, p):
+        return bytes([(n << 3) | 2, len(p)]) + p
+    assert b == (_ld(1, b"DevicePlugin") + _ld(2, b"amd.com/X")
+                 + _ld(3, b"/e.sock") + _ld(4, b"v1beta1"))
