@@ -114,7 +114,16 @@ class PluginManager:
     def _on_socket_removed(self, socket_name: str) -> None:
         for plugin in self.plugins.values():
             if plugin.socket_name == socket_name and not self._stop.is_set():
+                # Skip removals the plugin inflicted on itself (stop or
+                # restart); only an EXTERNAL wipe (kubelet cleanup) should
+                # trigger recovery.
+                if plugin.consume_expected_removal():
+                    log.info("socket removal of %s was self-inflicted; skipping", socket_name)
+                    continue
+                if os.path.exists(plugin.socket_path) and plugin.serving:
+                    continue
                 try:
+                    log.warning("EXTERNAL socket removal; restarting %s", plugin.resource_name)
                     plugin.restart()
                 except Exception:
                     log.exception("restart of %s failed", plugin.resource_name)

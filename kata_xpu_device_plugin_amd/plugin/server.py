@@ -92,6 +92,11 @@ class XPUDevicePlugin:
         # start/stop/restart may be invoked from the manager thread AND the
         # health watcher (socket-removal recovery) concurrently; serialize.
         self._lifecycle = threading.Lock()
+        # Self-initiated socket unlinks (stop/restart) raise inotify events
+        # indistinguishable from a kubelet wipe; count them so the watcher
+        # can tell recovery-worthy removals from our own.
+        self._removal_lock = threading.Lock()
+        self._expected_removals = 0
         self.watcher_servicer = None  # set in watcher registration mode
         self.watcher_socket_path: Optional[str] = None
         self.allocations = 0          # metrics
@@ -314,6 +319,7 @@ class XPUDevicePlugin:
                 options=[("grpc.max_concurrent_streams", 64)])
             api.add_device_plugin_servicer(server, self)
             if os.path.exists(self.socket_path):
+                self._note_expected_removal()
                 os.unlink(self.socket_path)
             server.add_insecure_port(f"unix://{self.socket_path}")
             if self.cfg.registration_mode in ("watcher", "both"):
@@ -380,6 +386,11 @@ class XPUDevicePlugin:
     def _stop_locked(self) -> None:
         self._stop.set()
         self._serving.clear()
+        # grpc core unlinks the unix socket file itself during server
+        # shutdown — mark the removal as self-inflicted BEFORE stopping so
+        # the health watcher doesn't treat it as a kubelet wipe.
+        if os.path.exists(self.socket_path):
+            self._note_expected_removal()
         if self._loop is not None and getattr(self, "_stop_async", None) is not None:
             try:
                 self._loop.call_soon_threadsafe(self._stop_async.set)
@@ -404,6 +415,18 @@ class XPUDevicePlugin:
         restarted plugin from global shutdown; our single Event does not)."""
         self.stop()
         self.start(register=register)
+
+    def _note_expected_removal(self) -> None:
+        with self._removal_lock:
+            self._expected_removals += 1
+
+    def consume_expected_removal(self) -> bool:
+        """True if the latest socket-removal event was self-inflicted."""
+        with self._removal_lock:
+            if self._expected_removals > 0:
+                self._expected_removals -= 1
+                return True
+            return False
 
     @property
     def serving(self) -> bool:
