@@ -25,6 +25,7 @@ log = get_logger(__name__)
 # other tenants, so floors are deliberately conservative.
 HBM_GBPS_FLOOR_MI355X = 4000.0
 BF16_TFLOPS_FLOOR_MI355X = 1000.0
+PCIE_GBPS_FLOOR = 20.0  # Gen5 x16 is ~63 GB/s; below 20 = degraded link
 
 
 def _ext():
@@ -46,6 +47,8 @@ class ProbeReport:
     compute_units: int = 0
     total_mem_gib: float = 0.0
     hbm_gbps: float = 0.0
+    pcie_h2d_gbps: float = 0.0
+    pcie_d2h_gbps: float = 0.0
     bf16_tflops: float = 0.0
     mfma_f32_ok: bool = False
     mfma_bf16_ok: bool = False
@@ -88,6 +91,16 @@ def probe_device(
             f"HBM bandwidth {rep.hbm_gbps:.0f} GB/s < floor {hbm_floor_gbps:.0f}"
         )
 
+    pcie = g.pcie_bandwidth_probe(dev, min(bandwidth_bytes, 256 << 20), 5)
+    rep.pcie_h2d_gbps = pcie["h2d_gbps"]
+    rep.pcie_d2h_gbps = pcie["d2h_gbps"]
+    for direction, v in (("H2D", rep.pcie_h2d_gbps), ("D2H", rep.pcie_d2h_gbps)):
+        if v < PCIE_GBPS_FLOOR:
+            rep.failures.append(
+                f"PCIe {direction} {v:.1f} GB/s < floor {PCIE_GBPS_FLOOR:.0f} "
+                "(degraded host link)"
+            )
+
     f32 = g.mfma_probe_f32(dev)
     rep.mfma_f32_ok = bool(f32["ok"])
     if not rep.mfma_f32_ok:
@@ -110,9 +123,11 @@ def probe_device(
 
     rep.passed = not rep.failures
     log.info(
-        "probe dev%d %s (%s, %d CUs, %.0f GiB): HBM %.0f GB/s, bf16 %.0f TF — %s",
+        "probe dev%d %s (%s, %d CUs, %.0f GiB): HBM %.0f GB/s, "
+        "PCIe %.0f/%.0f GB/s, bf16 %.0f TF — %s",
         dev, rep.name, rep.gcn_arch, rep.compute_units, rep.total_mem_gib,
-        rep.hbm_gbps, rep.bf16_tflops, "PASS" if rep.passed else rep.failures,
+        rep.hbm_gbps, rep.pcie_h2d_gbps, rep.pcie_d2h_gbps, rep.bf16_tflops,
+        "PASS" if rep.passed else rep.failures,
     )
     return rep
 

@@ -227,6 +227,50 @@ py::dict hbm_bandwidth_probe(int dev, size_t bytes, int iters) {
     return d;
 }
 
+py::dict pcie_bandwidth_probe(int dev, size_t bytes, int iters) {
+    // Host-link health: pinned-memory H2D/D2H streaming. MI355X host link
+    // is PCIe Gen5 x16 (63 GB/s spec); a degraded link (wrong slot width,
+    // retraining) shows up far below that.
+    HIP_CHECK(hipSetDevice(dev));
+    void* host = nullptr;
+    void* d = nullptr;
+    HIP_CHECK(hipHostMalloc(&host, bytes));
+    HIP_CHECK(hipMalloc(&d, bytes));
+    std::memset(host, 0x5a, bytes);
+    hipEvent_t t0, t1;
+    HIP_CHECK(hipEventCreate(&t0));
+    HIP_CHECK(hipEventCreate(&t1));
+    auto run = [&](bool h2d) {
+        // warmup
+        HIP_CHECK(hipMemcpyAsync(h2d ? d : host, h2d ? host : d, bytes,
+                                 h2d ? hipMemcpyHostToDevice : hipMemcpyDeviceToHost, 0));
+        HIP_CHECK(hipDeviceSynchronize());
+        float best = 1e30f;
+        for (int i = 0; i < iters; i++) {
+            HIP_CHECK(hipEventRecord(t0));
+            HIP_CHECK(hipMemcpyAsync(h2d ? d : host, h2d ? host : d, bytes,
+                                     h2d ? hipMemcpyHostToDevice : hipMemcpyDeviceToHost, 0));
+            HIP_CHECK(hipEventRecord(t1));
+            HIP_CHECK(hipEventSynchronize(t1));
+            float ms = 0;
+            HIP_CHECK(hipEventElapsedTime(&ms, t0, t1));
+            if (ms < best) best = ms;
+        }
+        return (double)bytes / (best * 1e6);
+    };
+    double h2d = run(true);
+    double d2h = run(false);
+    HIP_CHECK(hipEventDestroy(t0));
+    HIP_CHECK(hipEventDestroy(t1));
+    HIP_CHECK(hipFree(d));
+    HIP_CHECK(hipHostFree(host));
+    py::dict r;
+    r["h2d_gbps"] = h2d;
+    r["d2h_gbps"] = d2h;
+    r["bytes"] = (uint64_t)bytes;
+    return r;
+}
+
 py::dict memtest(int dev, size_t bytes) {
     HIP_CHECK(hipSetDevice(dev));
     size_t n = bytes / sizeof(uint64_t);
@@ -371,6 +415,8 @@ PYBIND11_MODULE(_gpuprobe, m) {
     m.def("hbm_bandwidth_probe", &hbm_bandwidth_probe, py::arg("dev") = 0,
           py::arg("bytes") = (size_t)1 << 31, py::arg("iters") = 5);
     m.def("memtest", &memtest, py::arg("dev") = 0, py::arg("bytes") = (size_t)1 << 31);
+    m.def("pcie_bandwidth_probe", &pcie_bandwidth_probe, py::arg("dev") = 0,
+          py::arg("bytes") = (size_t)256 << 20, py::arg("iters") = 5);
     m.def("mfma_probe_f32", &mfma_probe_f32, py::arg("dev") = 0);
     m.def("mfma_probe_bf16", &mfma_probe_bf16, py::arg("dev") = 0,
           py::arg("burn_iters") = 20000);

@@ -92,3 +92,10 @@ def test_ident_tool_runs():
 
     doc = collect(Config())
     assert doc["functions"], doc
+
+
+def test_pcie_bandwidth(gp):
+    r = gp.pcie_bandwidth_probe(0, 256 << 20, 5)
+    # Gen5 x16 spec 63 GB/s; require a sane link both directions
+    assert r["h2d_gbps"] > 20, r
+    assert r["d2h_gbps"] > 20, r
