@@ -232,8 +232,13 @@ done:
 // buckets; must mirror topology/hive.py::preferred_sets (tested for parity).
 // ---------------------------------------------------------------------------
 
-constexpr int64_t W_XGMI = 1000000;
-constexpr int64_t W_NUMA = 1000;
+// Tier-separated weights (mirror topology/hive.py): one xGMI pair beats
+// every possible NUMA pair even on a 1024-device node, etc.
+constexpr int64_t W_XGMI = 1000000000000LL;
+constexpr int64_t W_NUMA = 1000000;
+// Search-effort cap (mirror hive.py::MAX_SEARCH_NODES): past this the
+// selector keeps the concentration-greedy result of the first descent.
+constexpr long MAX_SEARCH_NODES = 50000;
 
 struct Bucket {
     std::string hive;
@@ -282,8 +287,12 @@ std::vector<std::string> select_preferred(
                   });
         buckets.push_back(std::move(kv.second));
     }
+    // Cap-DESCENDING: the first DFS path (max take from the biggest
+    // buckets) is the concentration optimum, so the branch-and-bound
+    // prunes nearly everything else; best-fit preference comes from the
+    // packing term, not search order (mirror of hive.py).
     std::sort(buckets.begin(), buckets.end(), [](const Bucket& a, const Bucket& b) {
-        if (a.ids.size() != b.ids.size()) return a.ids.size() < b.ids.size();
+        if (a.ids.size() != b.ids.size()) return a.ids.size() > b.ids.size();
         if (a.hive != b.hive) return a.hive < b.hive;
         return a.numa < b.numa;
     });
@@ -314,28 +323,30 @@ std::vector<std::string> select_preferred(
     std::map<std::pair<std::string, long>, int> hive_free;
     for (int i = 0; i < nb; i++) hive_free[pkey(i)] += caps[i];
 
+    // suffix max affinity and per-suffix remaining hive capacities, for
+    // the admissible upper bound (mirror of hive.py's branch-and-bound).
+    std::vector<int64_t> max_aff_suffix(nb + 1, 0);
+    for (int i = nb - 1; i >= 0; i--)
+        max_aff_suffix[i] = std::max(max_aff_suffix[i + 1], affinity[i]);
+    std::vector<std::map<std::string, int>> hive_cap_suffix(nb + 1);
+    std::vector<std::map<long, int>> numa_cap_suffix(nb + 1);
+    for (int i = nb - 1; i >= 0; i--) {
+        hive_cap_suffix[i] = hive_cap_suffix[i + 1];
+        numa_cap_suffix[i] = numa_cap_suffix[i + 1];
+        if (!buckets[i].hive.empty()) hive_cap_suffix[i][buckets[i].hive] += caps[i];
+        if (buckets[i].numa != -1) numa_cap_suffix[i][buckets[i].numa] += caps[i];
+    }
+
     std::vector<int> take(nb, 0), best_take;
     int64_t best_score = -1;
     int64_t best_pack = 0;
+    std::map<std::string, int> taken_hive;
+    std::map<long, int> taken_numa;
 
     auto bucket_pair = [&](int i, int c) -> int64_t {
         int64_t w = !buckets[i].hive.empty() ? W_XGMI
                     : buckets[i].numa != -1 ? W_NUMA : 0;
         return w * c * (c - 1) / 2;
-    };
-    auto cross = [&]() -> int64_t {
-        int64_t s = 0;
-        for (int i = 0; i < nb; i++) {
-            if (!take[i]) continue;
-            for (int j = i + 1; j < nb; j++) {
-                if (!take[j]) continue;
-                if (!buckets[i].hive.empty() && buckets[i].hive == buckets[j].hive)
-                    s += W_XGMI * take[i] * take[j];
-                else if (buckets[i].numa == buckets[j].numa && buckets[i].numa != -1)
-                    s += W_NUMA * take[i] * take[j];
-            }
-        }
-        return s;
     };
     auto packing = [&]() -> int64_t {
         std::map<std::pair<std::string, long>, int> taken;
@@ -345,25 +356,78 @@ std::vector<std::string> select_preferred(
         for (auto& kv : taken) p -= hive_free[kv.first] - kv.second;
         return p;
     };
+    // Admissible bound on future xGMI pairs: marginal gain of the j-th
+    // future device in hive h (holding t) is t+j; the sum of the `left`
+    // largest marginals over all remaining hives over-estimates every
+    // feasible placement (mirror of hive.py::xgmi_upper).
+    auto top_marginals = [](std::vector<int>& gains, int left) -> int64_t {
+        if ((int)gains.size() > left) {
+            std::partial_sort(gains.begin(), gains.begin() + left, gains.end(),
+                              std::greater<int>());
+            gains.resize(left);
+        }
+        int64_t sum = 0;
+        for (int g : gains) sum += g;
+        return sum;
+    };
+    auto xgmi_upper = [&](int i, int left) -> int64_t {
+        std::vector<int> gains;
+        for (auto& kv : hive_cap_suffix[i]) {
+            auto it = taken_hive.find(kv.first);
+            int t = it != taken_hive.end() ? it->second : 0;
+            for (int j = 0; j < kv.second; j++) gains.push_back(t + j);
+        }
+        return top_marginals(gains, left);
+    };
+    // same marginal-gain bound for the NUMA tier
+    auto numa_upper = [&](int i, int left) -> int64_t {
+        std::vector<int> gains;
+        for (auto& kv : numa_cap_suffix[i]) {
+            auto it = taken_numa.find(kv.first);
+            int t = it != taken_numa.end() ? it->second : 0;
+            for (int j = 0; j < kv.second; j++) gains.push_back(t + j);
+        }
+        return top_marginals(gains, left);
+    };
 
+    long nodes_visited = 0;
     std::function<void(int, int, int64_t)> dfs = [&](int i, int left, int64_t acc) {
+        if (++nodes_visited > MAX_SEARCH_NODES) return;  // keep best so far
         if (left == 0) {
-            int64_t total = acc + cross();
             int64_t pack = packing();
-            if (best_take.empty() || total > best_score ||
-                (total == best_score && pack > best_pack)) {
-                best_score = total;
+            if (best_take.empty() || acc > best_score ||
+                (acc == best_score && pack > best_pack)) {
+                best_score = acc;
                 best_pack = pack;
                 best_take = take;
             }
             return;
         }
         if (i >= nb || suffix[i] < left) return;
-        for (int c = std::min(caps[i], left); c >= 0; c--) {
-            take[i] = c;
-            dfs(i + 1, left - c, acc + bucket_pair(i, c) + affinity[i] * c);
+        if (!best_take.empty()) {
+            int64_t ub = acc + W_XGMI * xgmi_upper(i, left) +
+                         W_NUMA * numa_upper(i, left) +
+                         max_aff_suffix[i] * left;
+            if (ub < best_score || (ub == best_score && best_pack == 0)) return;
         }
-        take[i] = 0;
+        const std::string& hive = buckets[i].hive;
+        const long numa = buckets[i].numa;
+        for (int c = std::min(caps[i], left); c >= 0; c--) {
+            int64_t inc = bucket_pair(i, c) + affinity[i] * c;
+            if (c) {
+                if (!hive.empty()) inc += W_XGMI * c * taken_hive[hive];
+                if (numa != -1) inc += W_NUMA * c * taken_numa[numa];
+                take[i] = c;
+                if (!hive.empty()) taken_hive[hive] += c;
+                if (numa != -1) taken_numa[numa] += c;
+            }
+            dfs(i + 1, left - c, acc + inc);
+            if (c) {
+                take[i] = 0;
+                if (!hive.empty()) taken_hive[hive] -= c;
+                if (numa != -1) taken_numa[numa] -= c;
+            }
+        }
     };
     dfs(0, need, 0);
     if (best_take.empty() && need > 0) return {};
