@@ -122,3 +122,37 @@ def test_revalidate_groups_batched(tmp_path):
         devices_dir, bad, [0x1002], "vfio-pci")
     assert "vanished" in _native.revalidate_groups(
         devices_dir, [("70", ["0000:ff:00.0"])], [0x1002], "vfio-pci")
+
+
+def test_selector_vf_scale_bounded():
+    """64-VF symmetric node: selection must stay fast (bounded search) and
+    concentration-optimal at every size."""
+    import time
+    from kata_xpu_device_plugin_amd.topology.hive import GPUTopology, score_set
+
+    topo = GPUTopology(source="hint")
+    bdf_of = {}
+    ids = []
+    for h in range(8):
+        for v in range(8):
+            did = f"{100 + h * 8 + v}"
+            bdf = f"0000:{0x10 + h:02x}:02.{v}"
+            bdf_of[did] = bdf
+            topo.hive_of[bdf] = f"hive-{h + 1}"
+            topo.numa_of[bdf] = h // 4
+            ids.append(did)
+    from kata_xpu_device_plugin_amd.topology.hive import W_NUMA, W_XGMI
+    for k in (8, 16, 32, 64):
+        t0 = time.perf_counter()
+        pick = preferred_sets(topo, bdf_of, ids, [], k, use_native=True)
+        elapsed = time.perf_counter() - t0
+        assert len(pick) == k
+        assert elapsed < 0.2, f"selection at k={k} took {elapsed:.3f}s"
+        # xGMI tier must be concentration-optimal: ceil(k/8) hives
+        hives = {topo.hive(bdf_of[d]) for d in pick}
+        assert len(hives) == (k + 7) // 8
+        # score sanity: full hives contribute C(8,2) xGMI pairs each
+        s = score_set(topo, [bdf_of[d] for d in pick])
+        full, rem = divmod(k, 8)
+        expect_xgmi = full * 28 + rem * (rem - 1) // 2
+        assert s // W_XGMI == expect_xgmi
