@@ -11,7 +11,7 @@ without a cluster, and operators can validate a node with
 from __future__ import annotations
 
 import os
-from dataclasses import dataclass, field
+from dataclasses import dataclass
 from typing import Dict, List
 
 from .spec import ANNOTATION_ATTACH_PCI, ANNOTATION_BDF, CDISpec, parse_qualified_name, read_spec

@@ -7,8 +7,6 @@ discovery stats, enabled with --metrics-port.
 """
 from __future__ import annotations
 
-from typing import Optional
-
 from prometheus_client import CollectorRegistry, Gauge, start_http_server
 
 from .utils.log import get_logger

@@ -15,7 +15,6 @@ wire-format tests.
 """
 from __future__ import annotations
 
-import os
 from dataclasses import dataclass, field
 from typing import Dict, List
 

@@ -34,12 +34,7 @@ import grpc
 import grpc.aio
 
 from ..cdi.spec import ANNOTATION_PREFIX, qualified_name
-from ..config import (
-    Config,
-    STRATEGY_CDI_ANNOTATIONS,
-    STRATEGY_CDI_CRI,
-    STRATEGY_DEVICE_NODES,
-)
+from ..config import Config, STRATEGY_CDI_ANNOTATIONS, STRATEGY_CDI_CRI
 from ..discovery.sysfs import read_hex, read_link_base
 from ..topology.hive import GPUTopology, preferred_sets
 from ..utils.log import get_logger

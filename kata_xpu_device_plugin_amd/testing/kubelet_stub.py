@@ -11,7 +11,7 @@ from __future__ import annotations
 import os
 import threading
 from concurrent import futures
-from dataclasses import dataclass, field
+from dataclasses import dataclass
 from typing import Dict, List, Optional
 
 import grpc
