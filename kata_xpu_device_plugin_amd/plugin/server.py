@@ -305,9 +305,16 @@ class XPUDevicePlugin:
                  self.cfg.registration_mode)
 
     def _serve_thread(self) -> None:
+        from concurrent.futures import ThreadPoolExecutor
+
         loop = asyncio.new_event_loop()
         self._loop = loop
         asyncio.set_event_loop(loop)
+        # Dedicated bounded executor for the blocking health-queue waits of
+        # ListAndWatch streams (kubelet holds 1-2; don't share the process
+        # default executor with other plugins/components).
+        loop.set_default_executor(ThreadPoolExecutor(
+            max_workers=4, thread_name_prefix=f"kxdp-lw-{self.socket_name}"))
 
         async def _main():
             server = grpc.aio.server(
