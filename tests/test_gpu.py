@@ -99,3 +99,17 @@ def test_pcie_bandwidth(gp):
     # Gen5 x16 spec 63 GB/s; require a sane link both directions
     assert r["h2d_gbps"] > 20, r
     assert r["d2h_gbps"] > 20, r
+
+
+def test_lds_bandwidth(gp):
+    """ds_read_b128 streaming: ≈150 TB/s chip-wide per the microarch
+    tables; flag anything under 80 TB/s."""
+    r = gp.lds_bandwidth_probe(0, 100000)
+    assert r["tbps"] > 80, r
+
+
+def test_hbm_latency(gp):
+    """Dependent-chain HBM load latency ≈900 cycles ≈ 375-450 ns at
+    2.0-2.4 GHz; flag pathological latency (>1.5 µs)."""
+    r = gp.hbm_latency_probe(0, 1 << 30, 2000000)
+    assert 100 < r["latency_ns"] < 1500, r
