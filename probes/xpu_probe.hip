@@ -86,23 +86,40 @@ __global__ void pattern_check(const uint64_t* __restrict__ buf, size_t n,
 // keeps the loop LDS-bound (4 LDS cycles vs 2 VALU cycles per b128).
 // A sick CU/LDS array shows up as a chip-wide rate far below spec.
 // ---------------------------------------------------------------------------
+typedef float lds_f32x4 __attribute__((ext_vector_type(4)));
+
 __global__ void __launch_bounds__(256, 4)
 lds_read_burn(float* __restrict__ out, int iters) {
-    __shared__ float4 buf[2048];  // 32 KiB → 4 WGs/CU fit in 160 KiB
-    for (int i = threadIdx.x; i < 2048; i += 256)
-        buf[i] = make_float4((float)i, 1.f, 2.f, 3.f);
+    __shared__ lds_f32x4 buf[2048];  // 32 KiB → 4 WGs/CU fit in 160 KiB
+    for (int i = threadIdx.x; i < 2048; i += 256) {
+        lds_f32x4 v = {(float)i, 1.f, 2.f, 3.f};
+        buf[i] = v;
+    }
     __syncthreads();
+    // Inline-asm ds_read_b128 (volatile): plain C++ float4 reads get their
+    // unused components dead-code-eliminated into ds_read_b32, which
+    // measures the 128 B/clk narrow-read rate instead of the 256 B/clk
+    // b128 rate (§LDS table). One scalar accumulate per read keeps the
+    // loop LDS-bound (8×4 LDS cycles vs 8×2 VALU per batch); the explicit
+    // s_waitcnt is mandatory — hipcc pads nothing inside asm (§5.7).
+    unsigned base = (unsigned)(threadIdx.x & 255) * 16u;  // lane-contiguous 16-B slots
     float acc = 0.f;
-    int idx = threadIdx.x;
     for (int it = 0; it < iters; it++) {
-#pragma unroll
-        for (int j = 0; j < 8; j++) {
-            // consecutive lanes hit consecutive 16-B slots: conflict-free
-            // 4×16-lane groups for ds_read_b128 (§LDS lane-group table)
-            float4 v = buf[(idx + j * 256) & 2047];
-            acc += v.x;
-        }
-        idx += 1;
+        lds_f32x4 v0, v1, v2, v3, v4, v5, v6, v7;
+        asm volatile(
+            "ds_read_b128 %0, %8 offset:0\n\t"
+            "ds_read_b128 %1, %8 offset:4096\n\t"
+            "ds_read_b128 %2, %8 offset:8192\n\t"
+            "ds_read_b128 %3, %8 offset:12288\n\t"
+            "ds_read_b128 %4, %8 offset:16384\n\t"
+            "ds_read_b128 %5, %8 offset:20480\n\t"
+            "ds_read_b128 %6, %8 offset:24576\n\t"
+            "ds_read_b128 %7, %8 offset:28672\n\t"
+            "s_waitcnt lgkmcnt(0)"
+            : "=v"(v0), "=v"(v1), "=v"(v2), "=v"(v3),
+              "=v"(v4), "=v"(v5), "=v"(v6), "=v"(v7)
+            : "v"(base));
+        acc += v0.x + v1.x + v2.x + v3.x + v4.x + v5.x + v6.x + v7.x;
     }
     out[(size_t)blockIdx.x * blockDim.x + threadIdx.x] = acc;
 }
