@@ -1,0 +1,150 @@
+"""Property-based tests (hypothesis): selector invariants, CDI spec
+round-trips, naming sanitization, wire-format round-trips."""
+import string
+
+from hypothesis import given, settings, strategies as st
+
+from kata_xpu_device_plugin_amd.cdi.spec import (
+    CDIDeviceEntry,
+    CDISpec,
+    parse_qualified_name,
+    qualified_name,
+)
+from kata_xpu_device_plugin_amd.discovery.naming import sanitize
+from kata_xpu_device_plugin_amd.plugin import api
+from kata_xpu_device_plugin_amd.topology.hive import (
+    GPUTopology,
+    preferred_sets,
+    score_set,
+)
+
+# --- selector ---------------------------------------------------------------
+
+hives = st.integers(min_value=0, max_value=3)
+numas = st.integers(min_value=-1, max_value=1)
+
+
+@st.composite
+def topology_instances(draw):
+    n = draw(st.integers(min_value=1, max_value=12))
+    topo = GPUTopology(source="hint")
+    bdf_of = {}
+    for i in range(n):
+        bdf = f"0000:{10 + i:02x}:00.0"
+        did = str(100 + i)
+        bdf_of[did] = bdf
+        h = draw(hives)
+        if h:
+            topo.hive_of[bdf] = f"hive-{h}"
+        nm = draw(numas)
+        if nm >= 0:
+            topo.numa_of[bdf] = nm
+    return topo, bdf_of
+
+
+@given(topology_instances(), st.integers(min_value=0, max_value=14),
+       st.booleans())
+@settings(max_examples=200, deadline=None)
+def test_selector_invariants(inst, size, use_native):
+    topo, bdf_of = inst
+    ids = sorted(bdf_of)
+    pick = preferred_sets(topo, bdf_of, ids, [], size, use_native=use_native)
+    if 0 < size <= len(ids):
+        assert len(pick) == size
+        assert len(set(pick)) == size
+        assert set(pick) <= set(ids)
+    else:
+        assert pick == []
+
+
+@given(topology_instances(), st.data())
+@settings(max_examples=150, deadline=None)
+def test_selector_native_python_score_parity(inst, data):
+    topo, bdf_of = inst
+    ids = sorted(bdf_of)
+    size = data.draw(st.integers(min_value=1, max_value=len(ids)))
+    must = data.draw(st.lists(st.sampled_from(ids), max_size=min(2, size),
+                              unique=True))
+    nat = preferred_sets(topo, bdf_of, ids, must, size, use_native=True)
+    py = preferred_sets(topo, bdf_of, ids, must, size, use_native=False)
+    assert len(nat) == len(py)
+    if nat:
+        assert score_set(topo, [bdf_of[d] for d in nat]) == \
+            score_set(topo, [bdf_of[d] for d in py])
+        assert set(must) <= set(nat)
+
+
+@given(topology_instances(), st.data())
+@settings(max_examples=100, deadline=None)
+def test_selector_monotone_in_must(inst, data):
+    """Forcing a subset of the optimal pick never lowers the score."""
+    topo, bdf_of = inst
+    ids = sorted(bdf_of)
+    size = data.draw(st.integers(min_value=1, max_value=len(ids)))
+    free = preferred_sets(topo, bdf_of, ids, [], size)
+    if not free:
+        return
+    forced = data.draw(st.lists(st.sampled_from(free), max_size=size,
+                                unique=True))
+    pick = preferred_sets(topo, bdf_of, ids, forced, size)
+    assert score_set(topo, [bdf_of[d] for d in pick]) == \
+        score_set(topo, [bdf_of[d] for d in free])
+
+
+# --- CDI --------------------------------------------------------------------
+
+name_st = st.text(alphabet=string.ascii_letters + string.digits, min_size=1,
+                  max_size=12)
+
+
+@given(st.lists(name_st, min_size=1, max_size=8, unique=True))
+@settings(max_examples=100, deadline=None)
+def test_cdi_qualified_name_roundtrip(names):
+    for n in names:
+        qn = qualified_name("amd.com/gpu", n)
+        assert parse_qualified_name(qn) == ("amd.com/gpu", n)
+
+
+@given(st.lists(name_st, min_size=1, max_size=8, unique=True))
+@settings(max_examples=50, deadline=None)
+def test_cdi_spec_write_read_roundtrip(tmp_path_factory, names):
+    import tempfile
+    from kata_xpu_device_plugin_amd.cdi.spec import read_spec, write_spec
+    spec = CDISpec(kind="amd.com/gpu")
+    for n in names:
+        spec.devices.append(CDIDeviceEntry(
+            name=n, annotations={"attach-pci": "true", "bdf": "0000:0a:00.0"},
+            device_nodes=[f"/dev/vfio/{n}"]))
+    d = tempfile.mkdtemp(prefix="kxdp-prop-")
+    for fmt in ("yaml", "json"):
+        path = write_spec(spec, d, "s", fmt)
+        back = read_spec(path)
+        assert back.device_names() == spec.device_names()
+        assert [dv.device_nodes for dv in back.devices] == \
+            [dv.device_nodes for dv in spec.devices]
+
+
+# --- naming -----------------------------------------------------------------
+
+@given(st.text(max_size=40))
+@settings(max_examples=200, deadline=None)
+def test_sanitize_always_k8s_safe(raw):
+    out = sanitize(raw)
+    assert all(c.isalnum() or c == "_" for c in out)
+    assert out == out.upper()
+    assert not out.startswith("_") and not out.endswith("_")
+
+
+# --- wire format ------------------------------------------------------------
+
+@given(st.lists(name_st, min_size=0, max_size=6),
+       st.dictionaries(name_st, name_st, max_size=4))
+@settings(max_examples=100, deadline=None)
+def test_allocate_response_roundtrip(ids, envs):
+    resp = api.ContainerAllocateResponse(
+        envs=envs,
+        cdi_devices=[api.CDIDevice(name=f"amd.com/gpu={i}") for i in ids],
+    )
+    back = api.ContainerAllocateResponse.FromString(resp.SerializeToString())
+    assert dict(back.envs) == envs
+    assert [c.name for c in back.cdi_devices] == [f"amd.com/gpu={i}" for i in ids]
