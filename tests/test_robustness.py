@@ -207,3 +207,40 @@ def test_cdi_spec_self_heal(tmp_path):
         assert spec.device_names() == ["70", "71"]
     finally:
         mgr.stop()
+
+
+def test_mixed_pf_vf_node(tmp_path):
+    """Hybrid node: 4 whole GPUs vfio-bound for passthrough + 1 PF kept on
+    amdgpu whose 4 VFs are vfio-bound — two resources, correct split
+    (BASELINE.json config #5 shape)."""
+    node = make_mock_node(str(tmp_path), n_gpus=4, kfd=False, hint=False)
+    # PF on amdgpu (not schedulable itself)
+    node.add_gpu(MockGPU(bdf="0000:60:00.0", driver="amdgpu", iommu_group="50",
+                         sriov_totalvfs=4))
+    for k in range(4):
+        node.add_gpu(MockGPU(bdf=f"0000:60:02.{k}", device_id=0x75B3,
+                             iommu_group=str(110 + k), physfn_bdf="0000:60:00.0"))
+    cfg = node.config()
+    stub = KubeletStub(cfg.kubelet_socket_dir)
+    stub.start()
+    mgr = PluginManager(cfg)
+    mgr.setup()
+    mgr.start()
+    try:
+        regs = stub.wait_for_registration(2)
+        names = sorted(r.resource_name for r in regs)
+        assert names == ["amd.com/INSTINCT_MI355X", "amd.com/INSTINCT_MI355X_VF"]
+        vf_state = mgr.states["amd.com/INSTINCT_MI355X_VF"]
+        assert vf_state.device_ids() == ["110", "111", "112", "113"]
+        # PF group 50 is not schedulable anywhere
+        for st_ in mgr.states.values():
+            assert st_.device(str(50)) is None
+        # allocate a VF end-to-end
+        r = next(x for x in regs if x.resource_name.endswith("_VF"))
+        ps = stub.plugin_stub(r.endpoint)
+        resp = ps.Allocate(api.AllocateRequest(container_requests=[
+            api.ContainerAllocateRequest(devices_ids=["110"])]))
+        assert resp.container_responses[0].cdi_devices[0].name == "amd.com/gpu=110"
+    finally:
+        mgr.stop()
+        stub.stop()
