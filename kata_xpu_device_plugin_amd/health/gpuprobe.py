@@ -47,6 +47,8 @@ class ProbeReport:
     compute_units: int = 0
     total_mem_gib: float = 0.0
     hbm_gbps: float = 0.0
+    hbm_latency_ns: float = 0.0
+    lds_tbps: float = 0.0
     pcie_h2d_gbps: float = 0.0
     pcie_d2h_gbps: float = 0.0
     bf16_tflops: float = 0.0
@@ -91,6 +93,18 @@ def probe_device(
             f"HBM bandwidth {rep.hbm_gbps:.0f} GB/s < floor {hbm_floor_gbps:.0f}"
         )
 
+    lds = g.lds_bandwidth_probe(dev, max(2000, burn_iters))
+    rep.lds_tbps = lds["tbps"]
+    if is_mi355x and rep.lds_tbps < 80.0:
+        rep.failures.append(
+            f"LDS bandwidth {rep.lds_tbps:.0f} TB/s < floor 80 (b128 spec ≈150)")
+
+    lat = g.hbm_latency_probe(dev, min(bandwidth_bytes, 1 << 30), 500000)
+    rep.hbm_latency_ns = lat["latency_ns"]
+    if rep.hbm_latency_ns > 1500.0:
+        rep.failures.append(
+            f"HBM latency {rep.hbm_latency_ns:.0f} ns > 1500 (healthy ≈350)")
+
     pcie = g.pcie_bandwidth_probe(dev, min(bandwidth_bytes, 256 << 20), 5)
     rep.pcie_h2d_gbps = pcie["h2d_gbps"]
     rep.pcie_d2h_gbps = pcie["d2h_gbps"]
@@ -123,10 +137,11 @@ def probe_device(
 
     rep.passed = not rep.failures
     log.info(
-        "probe dev%d %s (%s, %d CUs, %.0f GiB): HBM %.0f GB/s, "
-        "PCIe %.0f/%.0f GB/s, bf16 %.0f TF — %s",
+        "probe dev%d %s (%s, %d CUs, %.0f GiB): HBM %.0f GB/s @%.0f ns, "
+        "LDS %.0f TB/s, PCIe %.0f/%.0f GB/s, bf16 %.0f TF — %s",
         dev, rep.name, rep.gcn_arch, rep.compute_units, rep.total_mem_gib,
-        rep.hbm_gbps, rep.pcie_h2d_gbps, rep.pcie_d2h_gbps, rep.bf16_tflops,
+        rep.hbm_gbps, rep.hbm_latency_ns, rep.lds_tbps,
+        rep.pcie_h2d_gbps, rep.pcie_d2h_gbps, rep.bf16_tflops,
         "PASS" if rep.passed else rep.failures,
     )
     return rep

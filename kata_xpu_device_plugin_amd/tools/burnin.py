@@ -21,6 +21,9 @@ def main(argv=None) -> int:
                    help="small buffers / short burn (smoke, not soak)")
     p.add_argument("--bandwidth-gib", type=float, default=2.0)
     p.add_argument("--memtest-gib", type=float, default=2.0)
+    p.add_argument("--soak-minutes", type=float, default=0.0,
+                   help="repeat the full suite for this long; any failing "
+                        "pass fails the soak (thermal/sustained validation)")
     args = p.parse_args(argv)
     configure("INFO")
 
@@ -32,6 +35,26 @@ def main(argv=None) -> int:
             bandwidth_bytes=int(args.bandwidth_gib * (1 << 30)),
             memtest_bytes=int(args.memtest_gib * (1 << 30)),
         )
+    if args.soak_minutes > 0:
+        import time
+        deadline = time.monotonic() + args.soak_minutes * 60
+        passes, failures = 0, 0
+        worst = []
+        while time.monotonic() < deadline:
+            reports = probe_all(**kw)
+            if not reports:
+                print("no GPUs visible", file=sys.stderr)
+                return 2
+            passes += 1
+            if not all(r.passed for r in reports):
+                failures += 1
+                worst = [r.as_dict() for r in reports if not r.passed]
+        summary = {"soak_minutes": args.soak_minutes, "passes": passes,
+                   "failing_passes": failures, "failures": worst}
+        json.dump(summary, sys.stdout, indent=2)
+        print()
+        return 0 if failures == 0 else 1
+
     reports = probe_all(**kw)
     if args.json:
         json.dump([r.as_dict() for r in reports], sys.stdout, indent=2)
