@@ -111,6 +111,14 @@ def load_topology(cfg: Config, inv: NodeInventory) -> GPUTopology:
     # NUMA comes from PCI sysfs — readable regardless of bound driver.
     for dev in inv.devices.values():
         topo.numa_of[dev.primary.bdf] = dev.numa_node
+    # SR-IOV VFs sit on their PF's die and fabric: inherit the PF's hive
+    # when known, else group VFs of one PF as their own pseudo-hive (they
+    # still share silicon — far better than treating them as unrelated).
+    for dev in inv.devices.values():
+        fn = dev.primary
+        if fn.is_vf and fn.bdf not in topo.hive_of:
+            pf = fn.physfn_bdf
+            topo.hive_of[fn.bdf] = topo.hive_of.get(pf) or f"pf-{pf}"
     return topo
 
 

@@ -153,3 +153,33 @@ def test_hint_file_parsing(tmp_path):
     assert topo.hive("0000:0a:00.0") == topo.hive("0000:12:00.0") == "hive-1"
     assert topo.hive("0000:1a:00.0") == "hive-2"
     assert topology_from_hint(str(tmp_path / "missing.json")) is None
+
+
+def test_vf_inherits_pf_hive(tmp_path):
+    """SR-IOV VFs share their PF's fabric: VFs of a hive-mapped PF join
+    that hive; VFs of an unmapped PF form a per-PF pseudo-hive, so a
+    multi-VF pod still prefers one physical GPU."""
+    from kata_xpu_device_plugin_amd.testing.mocknode import MockGPU
+
+    node = make_mock_node(str(tmp_path), n_gpus=2, driver="vfio-pci")
+    # PF0 (in hive via hint) with 2 VFs; PF1 not hive-mapped with 2 VFs
+    for k in range(2):
+        node.add_gpu(MockGPU(bdf=f"0000:0a:02.{k}", device_id=0x75B3,
+                             iommu_group=str(120 + k), physfn_bdf="0000:0a:00.0"))
+        node.add_gpu(MockGPU(bdf=f"0000:66:02.{k}", device_id=0x75B3,
+                             iommu_group=str(130 + k), physfn_bdf="0000:66:00.0"))
+    cfg = node.config()
+    inv = scan_node(cfg)
+    topo = load_topology(cfg, inv)
+    # PF 0000:0a:00.0 is in the hint's hive-1 → its VFs inherit it
+    pf_hive = topo.hive("0000:0a:00.0")
+    assert pf_hive
+    assert topo.hive("0000:0a:02.0") == topo.hive("0000:0a:02.1") == pf_hive
+    # unmapped PF → pseudo-hive shared by its VFs only
+    assert topo.hive("0000:66:02.0") == topo.hive("0000:66:02.1") == \
+        "pf-0000:66:00.0"
+    # placement: a 2-VF pod sticks to one PF
+    ids = ["120", "121", "130", "131"]
+    pick = preferred_allocation(topo, inv, ids, [], 2)
+    hives = {topo.hive(inv.devices[d].primary.bdf) for d in pick}
+    assert len(hives) == 1
