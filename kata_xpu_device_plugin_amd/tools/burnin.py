@@ -21,6 +21,8 @@ def main(argv=None) -> int:
                    help="small buffers / short burn (smoke, not soak)")
     p.add_argument("--bandwidth-gib", type=float, default=2.0)
     p.add_argument("--memtest-gib", type=float, default=2.0)
+    p.add_argument("--parallel", action="store_true",
+                   help="probe all GPUs concurrently (one process per GPU)")
     p.add_argument("--soak-minutes", type=float, default=0.0,
                    help="repeat the full suite for this long; any failing "
                         "pass fails the soak (thermal/sustained validation)")
@@ -41,7 +43,7 @@ def main(argv=None) -> int:
         passes, failures = 0, 0
         worst = []
         while time.monotonic() < deadline:
-            reports = probe_all(**kw)
+            reports = probe_all(parallel=args.parallel, **kw)
             if not reports:
                 print("no GPUs visible", file=sys.stderr)
                 return 2
@@ -55,7 +57,7 @@ def main(argv=None) -> int:
         print()
         return 0 if failures == 0 else 1
 
-    reports = probe_all(**kw)
+    reports = probe_all(parallel=args.parallel, **kw)
     if args.json:
         json.dump([r.as_dict() for r in reports], sys.stdout, indent=2)
         print()
