@@ -127,6 +127,9 @@ class Config:
     # should not send them, but a stale scheduler view can; defensive
     # default on). The reference allocates regardless of health.
     reject_unhealthy: bool = field(default_factory=lambda: _env_bool("REJECT_UNHEALTHY", True))
+    # Periodic in-daemon GPU probing of HIP-visible (amdgpu-bound) devices;
+    # 0 = off (the default — pure-VFIO nodes have no HIP-visible GPUs).
+    gpu_probe_interval_s: float = field(default_factory=lambda: _env_float("GPU_PROBE_S", 0.0))
 
     # --- observability ---
     metrics_port: int = field(default_factory=lambda: _env_int("METRICS_PORT", 0))  # 0 = off

@@ -39,6 +39,7 @@ class PluginManager:
         self._stop = threading.Event()
         self._metrics = None
         self._amdsmi = None
+        self._gpu_probe = None
 
     # ------------------------------------------------------------------
     def _group_by_resource(self, inv: NodeInventory) -> Dict[str, Dict[str, XPUDevice]]:
@@ -106,6 +107,12 @@ class PluginManager:
                 self.cfg.health_poll_interval_s, self._on_amdsmi_health
             )
             self._amdsmi.start()
+        if self.cfg.gpu_probe_interval_s > 0:
+            from ..health.probe_poller import GpuProbePoller
+            self._gpu_probe = GpuProbePoller(
+                self.cfg.gpu_probe_interval_s, self._on_probe_health
+            )
+            self._gpu_probe.start()
         if self.cfg.metrics_port:
             from ..metrics import MetricsExporter
             self._metrics = MetricsExporter(self)
@@ -152,6 +159,18 @@ class PluginManager:
                     state.set_health(gid, healthy, source="amdsmi")
                     return
 
+    def _on_probe_health(self, bdf: str, healthy: bool) -> None:
+        """In-daemon GPU probe verdict → DeviceState (source "probe")."""
+        bdf = bdf.lower()
+        for state in self.states.values():
+            for gid in state.device_ids():
+                dev = state.device(gid)
+                if dev and bdf in (fn.bdf.lower() for fn in dev.functions):
+                    if not healthy:
+                        log.warning("gpu probe: %s unhealthy", bdf)
+                    state.set_health(gid, healthy, source="probe")
+                    return
+
     def _on_kubelet_restarted(self) -> None:
         if self._stop.is_set():
             return
@@ -182,6 +201,9 @@ class PluginManager:
         if self._amdsmi is not None:
             self._amdsmi.stop()
             self._amdsmi = None
+        if self._gpu_probe is not None:
+            self._gpu_probe.stop()
+            self._gpu_probe = None
         for plugin in self.plugins.values():
             plugin.stop()
         if self._metrics is not None:

@@ -329,3 +329,43 @@ def test_watcher_wire_format():
         return bytes([(n << 3) | 2, len(p)]) + p
     assert b == (_ld(1, b"DevicePlugin") + _ld(2, b"amd.com/X")
                  + _ld(3, b"/e.sock") + _ld(4, b"v1beta1"))
+
+
+# --- in-daemon GPU probe poller (faked snapshots) --------------------------
+
+def test_gpu_probe_poller_transitions(tmp_path):
+    from kata_xpu_device_plugin_amd.health.probe_poller import GpuProbePoller
+    state = {"0000:0a:00.0": True}
+    calls = []
+    p = GpuProbePoller(999, lambda b, h: calls.append((b, h)),
+                       snapshot_fn=lambda: dict(state))
+    p.poll_once()
+    assert calls == []
+    state["0000:0a:00.0"] = False
+    p.poll_once()
+    p.poll_once()
+    assert calls == [("0000:0a:00.0", False)]
+    state["0000:0a:00.0"] = True
+    p.poll_once()
+    assert calls[-1] == ("0000:0a:00.0", True)
+
+
+def test_gpu_probe_marks_device_unhealthy(tmp_path):
+    """Probe verdict reaches DeviceState under the 'probe' source and
+    composes with the vfio source."""
+    node = make_mock_node(str(tmp_path), n_gpus=2, kfd=False, hint=False)
+    mgr = PluginManager(node.config())
+    mgr.setup()
+    st = mgr.states["amd.com/INSTINCT_MI355X"]
+    mgr._on_probe_health("0000:0A:00.0", False)  # case-insensitive match
+    assert not st.is_healthy("70")
+    st.set_health("70", False, source="vfio")
+    mgr._on_probe_health("0000:0a:00.0", True)
+    assert not st.is_healthy("70")  # vfio verdict still outstanding
+    st.set_health("70", True, source="vfio")
+    assert st.is_healthy("70")
+
+
+def test_probe_snapshot_no_gpu_is_empty():
+    from kata_xpu_device_plugin_amd.health.probe_poller import probe_snapshot
+    assert probe_snapshot() == {}  # CPU container: 0 HIP devices
