@@ -113,3 +113,17 @@ def test_hbm_latency(gp):
     2.0-2.4 GHz; flag pathological latency (>1.5 µs)."""
     r = gp.hbm_latency_probe(0, 1 << 30, 2000000)
     assert 100 < r["latency_ns"] < 1500, r
+
+
+def test_probe_snapshot_bdf_mapping(gp):
+    """HIP device → BDF mapping must land on a real sysfs AMD GPU function."""
+    from kata_xpu_device_plugin_amd.config import Config
+    from kata_xpu_device_plugin_amd.discovery.sysfs import scan_functions
+    from kata_xpu_device_plugin_amd.health.probe_poller import probe_snapshot
+
+    snap = probe_snapshot(quick_bytes=32 << 20, burn_iters=500)
+    assert snap, "at least one HIP-visible GPU expected"
+    assert all(snap.values()), snap
+    fns = {f.bdf for f in scan_functions(Config(sysfs_root="/sys")) if f.is_gpu}
+    for bdf in snap:
+        assert bdf in fns, (bdf, sorted(fns))
