@@ -85,3 +85,19 @@ def test_vf_bdfs_listing(tmp_path):
         os.symlink(os.path.join(node.sysfs, "bus", "pci", "devices", vf_bdf),
                    os.path.join(d, f"virtfn{k}"))
     assert vf_bdfs(cfg, pf.bdf) == ["0000:0a:02.0", "0000:0a:02.1"]
+
+
+def test_resourceslice_tool(tmp_path, monkeypatch, capsys):
+    import json as _json
+    from kata_xpu_device_plugin_amd.tools.resourceslice import main as rs_main
+    node = make_mock_node(str(tmp_path), n_gpus=4)
+    cfg = node.config()
+    monkeypatch.setenv("KXDP_SYSFS_ROOT", cfg.sysfs_root)
+    monkeypatch.setenv("KXDP_DEV_ROOT", cfg.dev_root)
+    monkeypatch.setenv("KXDP_TOPOLOGY_HINT", cfg.topology_hint_path)
+    rc = rs_main(["--node", "n1", "--pool", "p1"])
+    assert rc == 0
+    obj = _json.loads(capsys.readouterr().out)
+    assert obj["kind"] == "ResourceSlice"
+    assert obj["spec"]["nodeName"] == "n1"
+    assert len(obj["spec"]["devices"]) == 4
