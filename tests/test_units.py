@@ -369,3 +369,26 @@ def test_gpu_probe_marks_device_unhealthy(tmp_path):
 def test_probe_snapshot_no_gpu_is_empty():
     from kata_xpu_device_plugin_amd.health.probe_poller import probe_snapshot
     assert probe_snapshot() == {}  # CPU container: 0 HIP devices
+
+
+def test_registration_mode_both(tmp_path):
+    """'both' serves watcher registry socket AND self-registers legacy."""
+    import tempfile as _tf
+    node = make_mock_node(str(tmp_path), n_gpus=1)
+    registry = _tf.mkdtemp(prefix="kxdp-reg2-")
+    cfg = node.config(registration_mode="both", plugins_registry_dir=registry)
+    stub = KubeletStub(cfg.kubelet_socket_dir)
+    stub.start()
+    mgr = PluginManager(cfg)
+    mgr.setup()
+    mgr.start(register=True)
+    try:
+        regs = stub.wait_for_registration(1)  # legacy self-registration
+        assert regs[0].resource_name == "amd.com/INSTINCT_MI355X"
+        plugin = next(iter(mgr.plugins.values()))
+        assert os.path.exists(plugin.watcher_socket_path)  # watcher socket
+    finally:
+        mgr.stop()
+        stub.stop()
+        assert not os.path.exists(plugin.watcher_socket_path), \
+            "watcher socket must be cleaned up on stop"
