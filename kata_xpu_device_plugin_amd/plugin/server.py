@@ -312,9 +312,12 @@ class XPUDevicePlugin:
         asyncio.set_event_loop(loop)
         # Dedicated bounded executor for the blocking health-queue waits of
         # ListAndWatch streams (kubelet holds 1-2; don't share the process
-        # default executor with other plugins/components).
-        loop.set_default_executor(ThreadPoolExecutor(
-            max_workers=4, thread_name_prefix=f"kxdp-lw-{self.socket_name}"))
+        # default executor with other plugins/components). Shut down
+        # explicitly below: loop.close() does NOT stop executor threads,
+        # which would leak 4 threads per restart cycle.
+        lw_executor = ThreadPoolExecutor(
+            max_workers=4, thread_name_prefix=f"kxdp-lw-{self.socket_name}")
+        loop.set_default_executor(lw_executor)
 
         async def _main():
             server = grpc.aio.server(
@@ -355,6 +358,7 @@ class XPUDevicePlugin:
             self._start_error = e
             self._loop_ready.set()
         finally:
+            lw_executor.shutdown(wait=False, cancel_futures=True)
             try:
                 loop.close()
             except Exception:
