@@ -25,6 +25,8 @@ this build environment). The preparation logic and tests are real.
 """
 from __future__ import annotations
 
+import json
+import os
 import threading
 from typing import Dict, List, Optional
 
@@ -128,17 +130,42 @@ DRIVER_NAME = "gpu.amd.com"
 
 
 class ClaimStore:
-    """Prepared-claim bookkeeping (uid → device names). DRA drivers must
-    make NodePrepareResources idempotent and NodeUnprepareResources safe
-    for unknown claims."""
+    """Prepared-claim bookkeeping (uid → device names), optionally durable.
 
-    def __init__(self):
+    DRA semantics require NodePrepareResources to be idempotent,
+    NodeUnprepareResources to be safe for unknown claims, and prepared
+    state to SURVIVE a driver restart (kubelet may unprepare a claim the
+    old process prepared). With `path` set, every mutation is persisted
+    atomically as JSON and reloaded on construction.
+    """
+
+    def __init__(self, path: Optional[str] = None):
         self._lock = threading.Lock()
         self._claims: Dict[str, List[str]] = {}
+        self._path = path
+        if path and os.path.exists(path):
+            try:
+                with open(path) as f:
+                    raw = json.load(f)
+                self._claims = {str(k): [str(d) for d in v]
+                                for k, v in raw.items()}
+            except (OSError, ValueError) as e:
+                log.warning("unreadable claim store %s: %s (starting empty)",
+                            path, e)
+
+    def _persist_locked(self) -> None:
+        if not self._path:
+            return
+        os.makedirs(os.path.dirname(self._path) or ".", exist_ok=True)
+        tmp = self._path + ".tmp"
+        with open(tmp, "w") as f:
+            json.dump(self._claims, f)
+        os.replace(tmp, self._path)
 
     def put(self, uid: str, devices: List[str]) -> None:
         with self._lock:
             self._claims[uid] = list(devices)
+            self._persist_locked()
 
     def get(self, uid: str) -> Optional[List[str]]:
         with self._lock:
@@ -147,7 +174,10 @@ class ClaimStore:
 
     def pop(self, uid: str) -> Optional[List[str]]:
         with self._lock:
-            return self._claims.pop(uid, None)
+            v = self._claims.pop(uid, None)
+            if v is not None:
+                self._persist_locked()
+            return v
 
     def all(self) -> Dict[str, List[str]]:
         with self._lock:
@@ -166,12 +196,12 @@ class DRAServicer:
     """
 
     def __init__(self, cfg: Config, inventory: NodeInventory, pool_name: str,
-                 resolver=None):
+                 resolver=None, store_path: Optional[str] = None):
         self.cfg = cfg
         self.inventory = inventory
         self.pool_name = pool_name
         self.resolver = resolver
-        self.store = ClaimStore()
+        self.store = ClaimStore(store_path)
 
     def _prepare_one(self, claim) -> object:
         uid = claim.uid

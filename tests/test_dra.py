@@ -128,3 +128,29 @@ def test_resource_slice_shape(tmp_path):
     assert attrs["amd.com/xgmiHive"]["string"].startswith("hive-")
     assert attrs["amd.com/numaNode"]["int"] == 0
     assert attrs["amd.com/isSriovVf"]["bool"] is False
+
+
+def test_claim_store_survives_restart(tmp_path):
+    """DRA requirement: a restarted driver must still unprepare (and
+    idempotently re-prepare) claims the previous process prepared."""
+    path = str(tmp_path / "claims.json")
+    s1 = dra.ClaimStore(path)
+    s1.put("uid-a", ["70", "71"])
+    s1.put("uid-b", ["74"])
+    s1.pop("uid-b")
+    # "restart"
+    s2 = dra.ClaimStore(path)
+    assert s2.get("uid-a") == ["70", "71"]
+    assert s2.get("uid-b") is None
+    s2.pop("uid-a")
+    s3 = dra.ClaimStore(path)
+    assert s3.all() == {}
+
+
+def test_claim_store_corrupt_file_starts_empty(tmp_path):
+    path = tmp_path / "claims.json"
+    path.write_text("{not json")
+    s = dra.ClaimStore(str(path))
+    assert s.all() == {}
+    s.put("u", ["70"])
+    assert dra.ClaimStore(str(path)).get("u") == ["70"]
