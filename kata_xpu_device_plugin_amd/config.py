@@ -176,6 +176,13 @@ class Config:
         g.add_argument("--metrics-port", dest="metrics_port", type=int)
         g.add_argument("--log-level", dest="log_level")
         g.add_argument("--native", dest="native", choices=("auto", "require", "off"))
+        g.add_argument("--registration-mode", dest="registration_mode",
+                       choices=("legacy", "watcher", "both"))
+        g.add_argument("--plugins-registry", dest="plugins_registry_dir")
+        g.add_argument("--unified-resource", dest="unified_resource_name")
+        g.add_argument("--cdi-format", dest="cdi_format", choices=("yaml", "json"))
+        g.add_argument("--gpu-probe-interval", dest="gpu_probe_interval_s",
+                       type=float)
 
     @classmethod
     def from_args(cls, args: Optional[argparse.Namespace] = None) -> "Config":
