@@ -127,3 +127,18 @@ def test_probe_snapshot_bdf_mapping(gp):
     fns = {f.bdf for f in scan_functions(Config(sysfs_root="/sys")) if f.is_gpu}
     for bdf in snap:
         assert bdf in fns, (bdf, sorted(fns))
+
+
+def test_amdsmi_live_snapshot():
+    """amd-smi health snapshot must return real data for the amdgpu-bound
+    GPU (guards against silently-wrong amdsmi API names: every call is
+    defensively wrapped, so only a live check can notice)."""
+    from kata_xpu_device_plugin_amd.health.amdsmi_health import snapshot
+
+    snap = snapshot()
+    assert snap, "amd-smi saw no devices on a GPU box"
+    dh = next(iter(snap.values()))
+    assert dh.bdf.count(":") == 2
+    assert dh.healthy
+    assert dh.temperature_c is not None and 10 < dh.temperature_c < 105, (
+        "temperature missing — amdsmi temp API name/enum likely wrong")
