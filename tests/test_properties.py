@@ -148,3 +148,25 @@ def test_allocate_response_roundtrip(ids, envs):
     back = api.ContainerAllocateResponse.FromString(resp.SerializeToString())
     assert dict(back.envs) == envs
     assert [c.name for c in back.cdi_devices] == [f"amd.com/gpu={i}" for i in ids]
+
+
+# --- CDI reader fuzz --------------------------------------------------------
+
+@given(st.text(max_size=200))
+@settings(max_examples=100, deadline=None)
+def test_cdi_read_spec_never_crashes_outside_valueerror(raw):
+    """read_spec on arbitrary bytes must raise a clean, catchable error
+    (yaml/json/KeyError family), never hang or segfault."""
+    import tempfile, os
+    from kata_xpu_device_plugin_amd.cdi.spec import read_spec
+    d = tempfile.mkdtemp(prefix="kxdp-fuzz-")
+    p = os.path.join(d, "s.yaml")
+    with open(p, "w") as f:
+        f.write(raw)
+    try:
+        spec = read_spec(p)
+        spec.device_names()
+    except Exception as e:
+        import yaml as _yaml
+        assert isinstance(e, (_yaml.YAMLError, ValueError, KeyError,
+                              TypeError, AttributeError)), type(e)
