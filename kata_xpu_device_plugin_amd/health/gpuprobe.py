@@ -1,5 +1,9 @@
 """GPU burn-in / validation built on the _gpuprobe HIP extension.
 
+The reference has no vendor-library health at all (no
+NVML; fsnotify only, generic_device_plugin.go:389-457) — these probes are
+the MI355X-native additive capability SURVEY.md §2.2 calls for.
+
 Use cases for a passthrough device plugin:
 * pre-flight: before GPUs are vfio-bound (while still on amdgpu), verify
   each device's HBM bandwidth, matrix cores and memory integrity, and

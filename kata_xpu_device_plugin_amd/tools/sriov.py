@@ -1,5 +1,8 @@
 """SR-IOV (MxGPU) VF management.
 
+SURVEY.md §7 hard-parts item "SR-IOV/MxGPU VFs" (absent from the
+reference, which handles whole GPUs only).
+
 BASELINE.json config #5: MI355X SR-IOV VFs exposed as partitioned devices
 to Kata pods. This tool drives the standard sysfs SR-IOV interface:
 

@@ -1,5 +1,9 @@
 """Topology snapshot tool.
 
+MI355X-native analog of SURVEY.md §2.2's "xGMI-hive topology from
+sysfs" concept row (the reference has no topology awareness at all; its
+GetPreferredAllocation is a stub, generic_device_plugin.go:378-386).
+
 KFD only enumerates amdgpu-bound GPUs, so the xGMI hive layout must be
 captured BEFORE binding GPUs to vfio-pci. Run this once per node (e.g. an
 init container or provisioning step):
