@@ -299,10 +299,39 @@ class XPUDevicePlugin:
             ch.close()
         self._serving.set()
         if register and self.cfg.registration_mode in ("legacy", "both"):
-            self.register_with_kubelet()
+            # kubelet may not be up yet (DaemonSet starting before/with
+            # kubelet at node boot): a failed initial registration must not
+            # kill the plugin — serve anyway and retry in the background
+            # (the node watcher additionally re-registers on kubelet.sock
+            # creation). Reference behavior: Start() fails outright
+            # (generic_device_plugin.go:128-168).
+            try:
+                self.register_with_kubelet()
+            except Exception as e:
+                log.warning("kubelet registration failed (%s); serving "
+                            "anyway and retrying in background", e)
+                self._spawn_registration_retry()
         log.info("plugin %s serving on %s (%s registration)",
                  self.resource_name, self.socket_path,
                  self.cfg.registration_mode)
+
+    def _spawn_registration_retry(self) -> None:
+        def retry():
+            delay = 1.0
+            while self._serving.is_set() and not self._stop.is_set():
+                if self._stop.wait(delay):
+                    return
+                if not self._serving.is_set():
+                    return
+                try:
+                    self.register_with_kubelet()
+                    log.info("kubelet registration succeeded after retry")
+                    return
+                except Exception:
+                    delay = min(delay * 2, 30.0)
+
+        threading.Thread(target=retry, daemon=True,
+                         name=f"kxdp-register-retry-{self.socket_name}").start()
 
     def _serve_thread(self) -> None:
         from concurrent.futures import ThreadPoolExecutor

@@ -244,3 +244,27 @@ def test_mixed_pf_vf_node(tmp_path):
     finally:
         mgr.stop()
         stub.stop()
+
+
+def test_daemon_survives_missing_kubelet(tmp_path):
+    """kubelet not up at daemon start (node-boot race): the plugin must
+    serve anyway and register as soon as kubelet appears."""
+    node = make_mock_node(str(tmp_path), n_gpus=1)
+    cfg = node.config()
+    cfg.grpc_timeout_s = 0.5  # fail the initial dial fast
+    mgr = PluginManager(cfg)
+    mgr.setup()
+    mgr.start(register=True)  # no kubelet.sock exists yet
+    stub = None
+    try:
+        plugin = next(iter(mgr.plugins.values()))
+        assert plugin.serving, "plugin must serve despite missing kubelet"
+        # kubelet comes up late
+        stub = KubeletStub(cfg.kubelet_socket_dir)
+        stub.start()
+        regs = stub.wait_for_registration(1, timeout=15)
+        assert regs[0].resource_name == "amd.com/INSTINCT_MI355X"
+    finally:
+        mgr.stop()
+        if stub:
+            stub.stop()
