@@ -392,3 +392,26 @@ def test_registration_mode_both(tmp_path):
         stub.stop()
         assert not os.path.exists(plugin.watcher_socket_path), \
             "watcher socket must be cleaned up on stop"
+
+
+def test_json_log_format(capsys):
+    import json as _json
+    import logging
+    from kata_xpu_device_plugin_amd.utils import log as kxlog
+    # force a fresh handler with json format
+    root = logging.getLogger("kxdp")
+    for h in list(root.handlers):
+        root.removeHandler(h)
+    kxlog._configured = False
+    kxlog.configure("INFO", fmt="json")
+    kxlog.get_logger("kxdp.test").info("hello %s", "world")
+    err = capsys.readouterr().err.strip().splitlines()[-1]
+    doc = _json.loads(err)
+    assert doc["msg"] == "hello world"
+    assert doc["level"] == "INFO"
+    assert doc["logger"] == "kxdp.test"
+    # restore text format for other tests
+    for h in list(root.handlers):
+        root.removeHandler(h)
+    kxlog._configured = False
+    kxlog.configure("INFO")
