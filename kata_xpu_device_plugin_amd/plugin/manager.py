@@ -181,15 +181,21 @@ class PluginManager:
                 log.exception("re-registration of %s failed", plugin.resource_name)
 
     # ------------------------------------------------------------------
-    def run_forever(self) -> None:
-        """Block until SIGTERM/SIGINT (the reference blocks on a channel it
-        never closes and installs no signal handler)."""
+    def install_signal_handlers(self) -> None:
+        """Install SIGTERM/SIGINT → orderly-stop handlers. Call EARLY
+        (before setup/serve): a signal during startup must still shut the
+        daemon down cleanly, not kill it mid-registration."""
         def _sig(signum, frame):
             log.info("signal %d — shutting down", signum)
             self._stop.set()
 
         signal.signal(signal.SIGTERM, _sig)
         signal.signal(signal.SIGINT, _sig)
+
+    def run_forever(self) -> None:
+        """Block until SIGTERM/SIGINT (the reference blocks on a channel it
+        never closes and installs no signal handler)."""
+        self.install_signal_handlers()   # idempotent; EARLY install in main()
         self._stop.wait()
         self.stop()
 
@@ -242,6 +248,7 @@ def main(argv: Optional[List[str]] = None) -> int:
     cfg = Config.from_args(args)
     configure_logging(cfg.log_level)
     mgr = PluginManager(cfg)
+    mgr.install_signal_handlers()   # before serving: no startup kill window
     mgr.setup()
     mgr.start()
     mgr.run_forever()

@@ -268,3 +268,46 @@ def test_daemon_survives_missing_kubelet(tmp_path):
         mgr.stop()
         if stub:
             stub.stop()
+
+
+def test_daemon_main_sigterm(tmp_path):
+    """End-to-end daemon process: `python -m kata_xpu_device_plugin_amd`
+    starts against a mock node, serves, and shuts down cleanly on SIGTERM
+    (the reference has no signal handling — device_plugin.go:114)."""
+    import signal
+    import subprocess
+    import sys
+
+    node = make_mock_node(str(tmp_path), n_gpus=2)
+    cfg = node.config()
+    stub = KubeletStub(cfg.kubelet_socket_dir)
+    stub.start()
+    env = dict(os.environ,
+               KXDP_SYSFS_ROOT=cfg.sysfs_root,
+               KXDP_DEV_ROOT=cfg.dev_root,
+               KXDP_CDI_DIR=cfg.cdi_dir,
+               KXDP_KUBELET_DIR=cfg.kubelet_socket_dir,
+               KXDP_TOPOLOGY_HINT=cfg.topology_hint_path,
+               KXDP_AMDSMI_HEALTH="false",
+               PYTHONPATH=os.path.dirname(os.path.dirname(
+                   os.path.abspath(__file__))))
+    proc = subprocess.Popen(
+        [sys.executable, "-m", "kata_xpu_device_plugin_amd"],
+        env=env, stdout=subprocess.PIPE, stderr=subprocess.PIPE, text=True)
+    try:
+        regs = stub.wait_for_registration(1, timeout=20)
+        assert regs[0].resource_name == "amd.com/INSTINCT_MI355X"
+        ps = stub.plugin_stub(regs[0].endpoint)
+        resp = ps.Allocate(api.AllocateRequest(container_requests=[
+            api.ContainerAllocateRequest(devices_ids=["70"])]))
+        assert resp.container_responses[0].cdi_devices[0].name == "amd.com/gpu=70"
+        proc.send_signal(signal.SIGTERM)
+        rc = proc.wait(timeout=15)
+        assert rc == 0, proc.stderr.read()[-2000:]
+        # socket cleaned up on shutdown
+        assert not os.path.exists(
+            os.path.join(cfg.kubelet_socket_dir, regs[0].endpoint))
+    finally:
+        if proc.poll() is None:
+            proc.kill()
+        stub.stop()
