@@ -43,6 +43,11 @@ class MetricsExporter:
             "kxdp_last_allocate_seconds", "Duration of the last Allocate()",
             ["resource"], registry=self.registry,
         )
+        self.g_alloc_seconds = Gauge(
+            "kxdp_allocate_seconds_total",
+            "Cumulative server-side Allocate() handler time",
+            ["resource"], registry=self.registry,
+        )
         self.g_scan_wall = Gauge(
             "kxdp_discovery_seconds", "Wall time of the last discovery scan",
             registry=self.registry,
@@ -59,6 +64,7 @@ class MetricsExporter:
             self.g_allocations.labels(rname).set(plugin.allocations)
             self.g_alloc_failures.labels(rname).set(plugin.allocate_failures)
             self.g_last_alloc.labels(rname).set(plugin.last_allocate_s)
+            self.g_alloc_seconds.labels(rname).set(plugin.allocate_seconds_total)
 
     def start(self, port: int) -> None:
         self._server, _ = start_http_server(port, registry=self.registry)

@@ -97,6 +97,7 @@ class XPUDevicePlugin:
         self.allocations = 0          # metrics
         self.allocate_failures = 0
         self.last_allocate_s = 0.0
+        self.allocate_seconds_total = 0.0
         # hot-path precomputation
         self._devices_dir = os.path.join(cfg.sysfs_root, "bus", "pci", "devices")
         self._vendor_list = list(cfg.vendor_allowlist)
@@ -242,6 +243,7 @@ class XPUDevicePlugin:
             await context.abort(grpc.StatusCode.INVALID_ARGUMENT, str(e))
         self.allocations += 1
         self.last_allocate_s = time.perf_counter() - t0
+        self.allocate_seconds_total += self.last_allocate_s
         return response
 
     async def GetPreferredAllocation(self, request, context):

@@ -101,3 +101,17 @@ def test_resourceslice_tool(tmp_path, monkeypatch, capsys):
     assert obj["kind"] == "ResourceSlice"
     assert obj["spec"]["nodeName"] == "n1"
     assert len(obj["spec"]["devices"]) == 4
+
+
+def test_all_cli_entrypoints_have_help():
+    """Every tools CLI must at least render --help (import-time errors in
+    any tool would break provisioning scripts)."""
+    import subprocess
+    import sys
+    mods = ["topo", "burnin", "ident", "bind", "sriov", "validate",
+            "assignments", "resourceslice"]
+    for m in mods:
+        out = subprocess.run(
+            [sys.executable, "-m", f"kata_xpu_device_plugin_amd.tools.{m}",
+             "--help"], capture_output=True, text=True, timeout=60)
+        assert out.returncode == 0, (m, out.stderr[-500:])
