@@ -415,3 +415,17 @@ def test_json_log_format(capsys):
         root.removeHandler(h)
     kxlog._configured = False
     kxlog.configure("INFO")
+
+
+def test_inotify_rm_watch_and_path_of(tmp_path):
+    from kata_xpu_device_plugin_amd.utils import inotify
+    with inotify.Inotify() as ino:
+        wd = ino.add_watch(str(tmp_path), inotify.IN_CREATE)
+        assert ino.path_of(wd) == str(tmp_path)
+        ino.rm_watch(wd)
+        assert ino.path_of(wd) is None
+        (tmp_path / "x").write_text("")
+        # removed watch → no events for the path (IGNORED event may appear
+        # for the removed wd; just ensure no CREATE for "x")
+        evs = ino.read_events(timeout=0.3)
+        assert not any(e.name == "x" and e.created for e in evs)
