@@ -55,6 +55,17 @@ if __name__ == "__main__" or "setuptools" in sys.modules:
                 f"{PKG}.plugin", f"{PKG}.health", f"{PKG}.utils",
                 f"{PKG}.testing", f"{PKG}.tools",
             ],
+            entry_points={
+                "console_scripts": [
+                    "kata-xpu-device-plugin-amd=kata_xpu_device_plugin_amd.plugin.manager:main",
+                    "kxdp-topo=kata_xpu_device_plugin_amd.tools.topo:main",
+                    "kxdp-burnin=kata_xpu_device_plugin_amd.tools.burnin:main",
+                    "kxdp-bind=kata_xpu_device_plugin_amd.tools.bind:main",
+                    "kxdp-sriov=kata_xpu_device_plugin_amd.tools.sriov:main",
+                    "kxdp-validate=kata_xpu_device_plugin_amd.tools.validate:main",
+                    "kxdp-ident=kata_xpu_device_plugin_amd.tools.ident:main",
+                ],
+            },
             ext_modules=[
                 Extension(
                     f"{PKG}._native",
