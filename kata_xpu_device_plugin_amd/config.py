@@ -130,6 +130,9 @@ class Config:
     # Periodic in-daemon GPU probing of HIP-visible (amdgpu-bound) devices;
     # 0 = off (the default — pure-VFIO nodes have no HIP-visible GPUs).
     gpu_probe_interval_s: float = field(default_factory=lambda: _env_float("GPU_PROBE_S", 0.0))
+    # Periodic re-discovery (picks up SR-IOV VF count changes, rebinds);
+    # 0 = off. Rescans that find no change are no-ops (no CDI rewrite).
+    rescan_interval_s: float = field(default_factory=lambda: _env_float("RESCAN_S", 0.0))
 
     # --- observability ---
     metrics_port: int = field(default_factory=lambda: _env_int("METRICS_PORT", 0))  # 0 = off

@@ -40,6 +40,7 @@ class PluginManager:
         self._metrics = None
         self._amdsmi = None
         self._gpu_probe = None
+        self._rescan_thread = None
 
     # ------------------------------------------------------------------
     def _group_by_resource(self, inv: NodeInventory) -> Dict[str, Dict[str, XPUDevice]]:
@@ -113,6 +114,10 @@ class PluginManager:
                 self.cfg.gpu_probe_interval_s, self._on_probe_health
             )
             self._gpu_probe.start()
+        if self.cfg.rescan_interval_s > 0:
+            self._rescan_thread = threading.Thread(
+                target=self._rescan_loop, name="kxdp-rescan", daemon=True)
+            self._rescan_thread.start()
         if self.cfg.metrics_port:
             from ..metrics import MetricsExporter
             self._metrics = MetricsExporter(self)
@@ -158,6 +163,13 @@ class PluginManager:
                         log.warning("amd-smi: %s unhealthy: %s", bdf, reasons)
                     state.set_health(gid, healthy, source="amdsmi")
                     return
+
+    def _rescan_loop(self) -> None:
+        while not self._stop.wait(self.cfg.rescan_interval_s):
+            try:
+                self.rescan()
+            except Exception:
+                log.exception("periodic rescan failed")
 
     def _on_probe_health(self, bdf: str, healthy: bool) -> None:
         """In-daemon GPU probe verdict → DeviceState (source "probe")."""
@@ -217,10 +229,26 @@ class PluginManager:
             self._metrics = None
 
     # ------------------------------------------------------------------
-    def rescan(self) -> None:
+    @staticmethod
+    def _inventory_signature(inv: NodeInventory):
+        return {
+            gid: (tuple(dev.bdfs), dev.model_device_id, dev.numa_node,
+                  dev.is_vf)
+            for gid, dev in inv.devices.items()
+        }
+
+    def rescan(self) -> bool:
         """Re-run discovery and swap device sets in place (e.g. after VF
-        count changes); plugins keep serving."""
+        count changes); plugins keep serving. Returns True if anything
+        changed (no-change rescans skip the CDI rewrite and state swap)."""
         inv = scan_node(self.cfg)
+        if self.inventory is not None and \
+                self._inventory_signature(inv) == \
+                self._inventory_signature(self.inventory):
+            return False
+        log.info("rescan: inventory changed (%d → %d devices)",
+                 len(self.inventory.devices) if self.inventory else 0,
+                 len(inv.devices))
         self.inventory = inv
         spec = build_spec(inv, self.cfg.cdi_kind, self.cfg.dev_root)
         self.cdi_spec_path = write_spec(
@@ -234,6 +262,7 @@ class PluginManager:
         if new:
             log.warning("rescan found new resource name(s) %s — restart the "
                         "daemon to serve them", sorted(new))
+        return True
 
 
 def main(argv: Optional[List[str]] = None) -> int:

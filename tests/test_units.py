@@ -429,3 +429,33 @@ def test_inotify_rm_watch_and_path_of(tmp_path):
         # for the removed wd; just ensure no CREATE for "x")
         evs = ino.read_events(timeout=0.3)
         assert not any(e.name == "x" and e.created for e in evs)
+
+
+def test_rescan_no_change_is_noop(tmp_path):
+    node = make_mock_node(str(tmp_path), n_gpus=2, kfd=False, hint=False)
+    mgr = PluginManager(node.config())
+    mgr.setup()
+    path = mgr.cdi_spec_path
+    mtime = os.path.getmtime(path)
+    assert mgr.rescan() is False          # nothing changed
+    assert os.path.getmtime(path) == mtime  # CDI not rewritten
+    node.add_gpu(MockGPU(bdf="0000:77:00.0", iommu_group="88"))
+    assert mgr.rescan() is True
+    assert "88" in mgr.states["amd.com/INSTINCT_MI355X"].device_ids()
+
+
+def test_periodic_rescan_thread(tmp_path):
+    node = make_mock_node(str(tmp_path), n_gpus=1, kfd=False, hint=False)
+    cfg = node.config(rescan_interval_s=0.1)
+    mgr = PluginManager(cfg)
+    mgr.setup()
+    mgr.start(register=False)
+    try:
+        node.add_gpu(MockGPU(bdf="0000:77:00.0", iommu_group="88"))
+        deadline = time.monotonic() + 5
+        st = mgr.states["amd.com/INSTINCT_MI355X"]
+        while "88" not in st.device_ids() and time.monotonic() < deadline:
+            time.sleep(0.05)
+        assert "88" in st.device_ids(), "periodic rescan must pick up new VF"
+    finally:
+        mgr.stop()
