@@ -64,6 +64,7 @@ if __name__ == "__main__" or "setuptools" in sys.modules:
                     "kxdp-sriov=kata_xpu_device_plugin_amd.tools.sriov:main",
                     "kxdp-validate=kata_xpu_device_plugin_amd.tools.validate:main",
                     "kxdp-ident=kata_xpu_device_plugin_amd.tools.ident:main",
+                    "kxdp-doctor=kata_xpu_device_plugin_amd.tools.doctor:main",
                 ],
             },
             ext_modules=[

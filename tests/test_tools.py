@@ -5,7 +5,7 @@ import os
 import pytest
 
 from kata_xpu_device_plugin_amd.config import Config
-from kata_xpu_device_plugin_amd.testing.mocknode import MockGPU, make_mock_node
+from kata_xpu_device_plugin_amd.testing.mocknode import MockGPU, MockNode, make_mock_node
 from kata_xpu_device_plugin_amd.tools.bind import Binder
 from kata_xpu_device_plugin_amd.tools.ident import collect
 from kata_xpu_device_plugin_amd.tools.sriov import set_numvfs, vf_bdfs
@@ -109,9 +109,58 @@ def test_all_cli_entrypoints_have_help():
     import subprocess
     import sys
     mods = ["topo", "burnin", "ident", "bind", "sriov", "validate",
-            "assignments", "resourceslice"]
+            "assignments", "resourceslice", "doctor"]
     for m in mods:
         out = subprocess.run(
             [sys.executable, "-m", f"kata_xpu_device_plugin_amd.tools.{m}",
              "--help"], capture_output=True, text=True, timeout=60)
         assert out.returncode == 0, (m, out.stderr[-500:])
+
+
+def _doctor_env(monkeypatch, cfg):
+    monkeypatch.setenv("KXDP_SYSFS_ROOT", cfg.sysfs_root)
+    monkeypatch.setenv("KXDP_DEV_ROOT", cfg.dev_root)
+    monkeypatch.setenv("KXDP_CDI_DIR", cfg.cdi_dir)
+    monkeypatch.setenv("KXDP_TOPOLOGY_HINT", cfg.topology_hint_path)
+
+
+def test_doctor_ready_node(tmp_path, monkeypatch, capsys):
+    from kata_xpu_device_plugin_amd.tools.doctor import main as doctor_main
+    node = make_mock_node(str(tmp_path), n_gpus=4)
+    _doctor_env(monkeypatch, node.config())
+    rc = doctor_main([])
+    out = capsys.readouterr().out
+    assert rc == 0, out
+    assert "verdict: OK" in out
+
+
+def test_doctor_detects_missing_vfio_node(tmp_path, monkeypatch, capsys):
+    from kata_xpu_device_plugin_amd.tools.doctor import main as doctor_main
+    node = make_mock_node(str(tmp_path), n_gpus=2)
+    node.remove_vfio_node("71")
+    _doctor_env(monkeypatch, node.config())
+    rc = doctor_main(["--json"])
+    import json as _json
+    doc = _json.loads(capsys.readouterr().out)
+    assert rc == 1
+    assert any("71" in p for p in doc["problems"])
+
+
+def test_doctor_preprovisioning_note(tmp_path, monkeypatch, capsys):
+    from kata_xpu_device_plugin_amd.tools.doctor import main as doctor_main
+    node = make_mock_node(str(tmp_path), n_gpus=2, driver="amdgpu",
+                          kfd=True, hint=False)
+    _doctor_env(monkeypatch, node.config())
+    rc = doctor_main([])
+    out = capsys.readouterr().out
+    assert rc == 0, out
+    assert "pre-provisioning" in out
+
+
+def test_doctor_no_gpus(tmp_path, monkeypatch, capsys):
+    from kata_xpu_device_plugin_amd.tools.doctor import main as doctor_main
+    node = MockNode(root=str(tmp_path))  # empty node
+    os.makedirs(os.path.join(node.sysfs, "bus", "pci", "devices"), exist_ok=True)
+    _doctor_env(monkeypatch, node.config())
+    rc = doctor_main([])
+    assert rc == 2
