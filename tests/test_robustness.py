@@ -362,3 +362,55 @@ def test_shutdown_under_load(tmp_path):
     stub.stop()
     assert not errs
     assert elapsed < 10, f"shutdown took {elapsed:.1f}s"
+
+
+def test_watcher_mode_kubelet_flow(tmp_path):
+    """Modern-kubelet flow end-to-end: watch the plugins_registry dir,
+    discover the registration socket, GetInfo → dial the advertised
+    endpoint → full DevicePlugin round-trip → NotifyRegistrationStatus."""
+    import tempfile
+    from kata_xpu_device_plugin_amd.plugin.watcher_registration import (
+        InfoRequest, RegistrationStatus, WatcherRegistrationStub)
+    from kata_xpu_device_plugin_amd.utils import inotify
+
+    node = make_mock_node(str(tmp_path), n_gpus=2)
+    registry = tempfile.mkdtemp(prefix="kxdp-reg3-")
+    cfg = node.config(registration_mode="watcher",
+                     plugins_registry_dir=registry)
+    ino = inotify.Inotify()
+    ino.add_watch(registry, inotify.IN_CREATE)
+    mgr = PluginManager(cfg)
+    mgr.setup()
+    mgr.start(register=True)
+    try:
+        # kubelet's plugin watcher sees the socket appear
+        deadline = time.monotonic() + 5
+        seen = set()
+        while time.monotonic() < deadline and not seen:
+            seen = {e.name for e in ino.read_events(timeout=0.5) if e.created}
+        plugin = next(iter(mgr.plugins.values()))
+        assert plugin.socket_name in seen, seen
+        # GetInfo on the registry socket
+        ch = grpc.insecure_channel(
+            f"unix://{os.path.join(registry, plugin.socket_name)}")
+        grpc.channel_ready_future(ch).result(timeout=5)
+        info = WatcherRegistrationStub(ch).GetInfo(InfoRequest())
+        assert info.type == "DevicePlugin"
+        # dial the advertised endpoint, full DevicePlugin round-trip
+        ch2 = grpc.insecure_channel(f"unix://{info.endpoint}")
+        grpc.channel_ready_future(ch2).result(timeout=5)
+        ps = api.DevicePluginStub(ch2)
+        stream = ps.ListAndWatch(api.Empty())
+        assert len(next(stream).devices) == 2
+        stream.cancel()
+        resp = ps.Allocate(api.AllocateRequest(container_requests=[
+            api.ContainerAllocateRequest(devices_ids=["70"])]))
+        assert resp.container_responses[0].cdi_devices[0].name == "amd.com/gpu=70"
+        WatcherRegistrationStub(ch).NotifyRegistrationStatus(
+            RegistrationStatus(plugin_registered=True))
+        assert plugin.watcher_servicer.last_status == (True, "")
+        ch.close()
+        ch2.close()
+    finally:
+        ino.close()
+        mgr.stop()

@@ -164,3 +164,26 @@ def test_doctor_no_gpus(tmp_path, monkeypatch, capsys):
     _doctor_env(monkeypatch, node.config())
     rc = doctor_main([])
     assert rc == 2
+
+
+def test_daemonset_manifest_sane():
+    """Guard the deploy manifest: parses, mounts every path the daemon
+    needs, and the container command points at this package."""
+    import yaml as _yaml
+    path = os.path.join(os.path.dirname(os.path.dirname(
+        os.path.abspath(__file__))), "deploy",
+        "kata-xpu-device-plugin-amd.yaml")
+    doc = _yaml.safe_load(open(path))
+    assert doc["kind"] == "DaemonSet"
+    spec = doc["spec"]["template"]["spec"]
+    ctr = spec["containers"][0]
+    assert "kata_xpu_device_plugin_amd" in " ".join(ctr["command"])
+    mounts = {m["mountPath"] for m in ctr["volumeMounts"]}
+    for required in ("/var/lib/kubelet/device-plugins", "/dev/vfio",
+                     "/var/run/cdi", "/sys",
+                     "/var/lib/kubelet/pod-resources",
+                     "/var/lib/kubelet/plugins_registry"):
+        assert required in mounts, required
+    vols = {v["name"] for v in spec["volumes"]}
+    assert {m["name"] for m in ctr["volumeMounts"]} <= vols
+    assert ctr["livenessProbe"]["httpGet"]["path"] == "/metrics"
