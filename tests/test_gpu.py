@@ -142,3 +142,16 @@ def test_amdsmi_live_snapshot():
     assert dh.healthy
     assert dh.temperature_c is not None and 10 < dh.temperature_c < 105, (
         "temperature missing — amdsmi temp API name/enum likely wrong")
+
+
+def test_doctor_live_node():
+    """kxdp-doctor on the live box: amdgpu-bound GPUs → pre-provisioning
+    verdict (exit 0) with real discovery/topology/amd-smi content."""
+    from kata_xpu_device_plugin_amd.config import Config
+    from kata_xpu_device_plugin_amd.tools.doctor import diagnose
+
+    doc = diagnose(Config())
+    assert not doc["problems"], doc["problems"]
+    assert doc["discovery"]["functions"] >= 1
+    assert any(g["driver"] == "amdgpu" for g in doc["gpus"])
+    assert doc["amdsmi"], "amd-smi must see the amdgpu-bound GPU"
