@@ -1,7 +1,7 @@
 PYTHON ?= python3
 IMAGE  ?= ghcr.io/example/kata-xpu-device-plugin-amd:0.1.0
 
-.PHONY: build build-hip test test-gpu bench image clean
+.PHONY: build build-hip test test-gpu bench image clean doctor validate burnin
 
 build:
 	$(PYTHON) setup.py build_ext --inplace
@@ -17,6 +17,15 @@ test-gpu:
 
 bench:
 	$(PYTHON) bench.py --steps 30 --warmup 5
+
+doctor:
+	$(PYTHON) -m kata_xpu_device_plugin_amd.tools.doctor
+
+validate:
+	$(PYTHON) -m kata_xpu_device_plugin_amd.tools.validate
+
+burnin:
+	$(PYTHON) -m kata_xpu_device_plugin_amd.tools.burnin
 
 image:
 	docker build -t $(IMAGE) .
