@@ -18,6 +18,9 @@ RUN python3 setup.py build_ext --inplace && \
     python3 -c "from setup import build_hip; build_hip('gfx950')"
 
 FROM python:3.10-slim
+# Control-plane runtime deps only. Optional extras on the node image:
+#   - amdsmi python bindings (ship with ROCm) for the amd-smi health poller
+#   - ROCm runtime libs if tools/burnin probes are used in-container
 RUN pip install --no-cache-dir grpcio protobuf pyyaml prometheus_client
 # pci.ids fallback for naming unknown silicon (reference side-loads it to
 # /usr/pci.ids, Dockerfile:66; we use the distro location first)
