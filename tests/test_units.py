@@ -459,3 +459,20 @@ def test_periodic_rescan_thread(tmp_path):
         assert "88" in st.device_ids(), "periodic rescan must pick up new VF"
     finally:
         mgr.stop()
+
+
+def test_state_watcher_queue_overflow_is_safe(tmp_path):
+    """A slow ListAndWatch consumer must never block or crash health
+    updates: past the queue bound, notifications are dropped (the stream
+    resyncs from a full snapshot on its next wakeup)."""
+    st, inv = _state_of(tmp_path, n=2)
+    q = st.watch()
+    for i in range(200):  # far beyond the 64-entry bound
+        st.set_health("70", i % 2 == 0)
+    assert q.qsize() <= 64
+    # consumer drains and sees a consistent final snapshot
+    while not q.empty():
+        q.get_nowait()
+    snap = dict((d.id, h) for d, h in st.snapshot())
+    assert snap["70"] in (True, False)
+    st.unwatch(q)
