@@ -65,6 +65,8 @@ if __name__ == "__main__" or "setuptools" in sys.modules:
                     "kxdp-validate=kata_xpu_device_plugin_amd.tools.validate:main",
                     "kxdp-ident=kata_xpu_device_plugin_amd.tools.ident:main",
                     "kxdp-doctor=kata_xpu_device_plugin_amd.tools.doctor:main",
+                    "kxdp-assignments=kata_xpu_device_plugin_amd.tools.assignments:main",
+                    "kxdp-resourceslice=kata_xpu_device_plugin_amd.tools.resourceslice:main",
                 ],
             },
             ext_modules=[
