@@ -8,6 +8,7 @@ discovery stats, enabled with --metrics-port.
 from __future__ import annotations
 
 from prometheus_client import CollectorRegistry, Gauge, start_http_server
+from prometheus_client.core import CounterMetricFamily
 
 from .utils.log import get_logger
 
@@ -21,6 +22,10 @@ class MetricsExporter:
         self._server = None
         # Registered FIRST so each scrape refreshes the gauges before the
         # registry collects them (collectors run in registration order).
+        # The *_total series are Counter-typed and emitted by the collector
+        # itself (monotonic cumulative values read straight off the plugin
+        # objects) so rate()/increase() reset-handling works — a Gauge-typed
+        # *_total violates Prometheus conventions (advisor finding r1).
         self.registry.register(_Refresher(self))
 
         self.g_devices = Gauge(
@@ -31,21 +36,8 @@ class MetricsExporter:
             "kxdp_devices_healthy", "Healthy xPU devices",
             ["resource"], registry=self.registry,
         )
-        self.g_allocations = Gauge(
-            "kxdp_allocations_total", "Allocate() calls served",
-            ["resource"], registry=self.registry,
-        )
-        self.g_alloc_failures = Gauge(
-            "kxdp_allocation_failures_total", "Allocate() calls rejected",
-            ["resource"], registry=self.registry,
-        )
         self.g_last_alloc = Gauge(
             "kxdp_last_allocate_seconds", "Duration of the last Allocate()",
-            ["resource"], registry=self.registry,
-        )
-        self.g_alloc_seconds = Gauge(
-            "kxdp_allocate_seconds_total",
-            "Cumulative server-side Allocate() handler time",
             ["resource"], registry=self.registry,
         )
         self.g_scan_wall = Gauge(
@@ -61,10 +53,24 @@ class MetricsExporter:
             state = m.states[rname]
             self.g_devices.labels(rname).set(len(state.device_ids()))
             self.g_healthy.labels(rname).set(len(state.healthy_ids()))
-            self.g_allocations.labels(rname).set(plugin.allocations)
-            self.g_alloc_failures.labels(rname).set(plugin.allocate_failures)
             self.g_last_alloc.labels(rname).set(plugin.last_allocate_s)
-            self.g_alloc_seconds.labels(rname).set(plugin.allocate_seconds_total)
+
+    def counter_families(self):
+        """Counter-typed *_total families, one sample per resource."""
+        c_allocs = CounterMetricFamily(
+            "kxdp_allocations", "Allocate() calls served", labels=["resource"])
+        c_fail = CounterMetricFamily(
+            "kxdp_allocation_failures", "Allocate() calls rejected",
+            labels=["resource"])
+        c_secs = CounterMetricFamily(
+            "kxdp_allocate_seconds",
+            "Cumulative server-side Allocate() handler time",
+            labels=["resource"])
+        for rname, plugin in self.manager.plugins.items():
+            c_allocs.add_metric([rname], plugin.allocations)
+            c_fail.add_metric([rname], plugin.allocate_failures)
+            c_secs.add_metric([rname], plugin.allocate_seconds_total)
+        return [c_allocs, c_fail, c_secs]
 
     def start(self, port: int) -> None:
         self._server, _ = start_http_server(port, registry=self.registry)
@@ -77,7 +83,8 @@ class MetricsExporter:
 
 
 class _Refresher:
-    """Collector that refreshes gauges before each scrape."""
+    """Collector that refreshes gauges before each scrape and emits the
+    Counter-typed *_total families."""
 
     def __init__(self, exporter: MetricsExporter):
         self.exporter = exporter
@@ -85,6 +92,7 @@ class _Refresher:
     def collect(self):
         try:
             self.exporter.refresh()
+            return self.exporter.counter_families()
         except Exception:
             log.exception("metrics refresh failed")
-        return []
+            return []

@@ -207,6 +207,15 @@ class XPUDevicePlugin:
         resp = api.ContainerAllocateResponse()
         strategy = self.cfg.device_list_strategy
         bdfs: List[str] = []
+        if ids and strategy not in (STRATEGY_CDI_CRI, STRATEGY_CDI_ANNOTATIONS):
+            # Raw device-nodes strategy: a plain (runc) container needs the
+            # VFIO container node /dev/vfio/vfio to open any group fd; CDI
+            # strategies leave this to the spec/runtime (Kata cold-plugs the
+            # PCI device and needs neither inside the host container).
+            vfio_ctl = os.path.join(self.cfg.dev_root, "vfio", "vfio")
+            resp.devices.add(
+                container_path=vfio_ctl, host_path=vfio_ctl, permissions="rw"
+            )
         for gid in ids:
             dev = self.state.device(gid)
             assert dev is not None  # validated earlier

@@ -214,8 +214,12 @@ def test_allocate_strategies(tmp_path, strategy, check):
             assert cr.annotations["cdi.k8s.io/vfio70"] == "amd.com/gpu=70"
             assert len(cr.cdi_devices) == 0
         else:
-            assert cr.devices[0].host_path.endswith("/vfio/70")
-            assert cr.devices[0].permissions == "rw"
+            # First node is the VFIO container device /dev/vfio/vfio (a runc
+            # container cannot open a group fd without it), then the group.
+            assert cr.devices[0].host_path.endswith("/vfio/vfio")
+            assert cr.devices[1].host_path.endswith("/vfio/70")
+            assert all(d.permissions == "rw" for d in cr.devices)
+            assert len(cr.devices) == 2
             assert len(cr.cdi_devices) == 0
             assert "KUBERNETES_CDI_VENDOR_CLASS" not in cr.envs
     finally:
@@ -261,6 +265,14 @@ def test_metrics_exporter_refresh(tmp_path):
     text = generate_latest(exp.registry).decode()
     assert 'kxdp_devices{resource="amd.com/INSTINCT_MI355X"} 2.0' in text
     assert "kxdp_discovery_seconds" in text
+    # *_total series must be Counter-typed (advisor r1): rate()/increase()
+    # depend on the counter contract, and the exposition name gets the
+    # _total suffix appended by the client library.
+    assert "# TYPE kxdp_allocations_total counter" in text
+    assert "# TYPE kxdp_allocation_failures_total counter" in text
+    assert "# TYPE kxdp_allocate_seconds_total counter" in text
+    assert 'kxdp_allocations_total{resource="amd.com/INSTINCT_MI355X"} 0.0' in text
+    assert "# TYPE kxdp_devices gauge" in text
 
 
 def test_state_multi_source_health(tmp_path):
