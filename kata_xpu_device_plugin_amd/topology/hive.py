@@ -307,8 +307,11 @@ def preferred_sets(
     def dfs(i: int, left: int, acc: int):
         nonlocal best_score, best_packing, best_take, nodes_visited
         nodes_visited += 1
-        if nodes_visited > MAX_SEARCH_NODES:
-            return  # budget exhausted: keep the best found so far
+        if nodes_visited > MAX_SEARCH_NODES and best_take is not None:
+            # Budget exhausted: keep the best found so far. The first
+            # greedy descent (≤ n_b nodes) always completes regardless of
+            # the cap, so a feasible request never degrades to [].
+            return
         if left == 0:
             p = packing(take)
             if acc > best_score or (acc == best_score and p > best_packing) \

@@ -74,6 +74,69 @@ def test_selector_native_python_score_parity(inst, data):
         assert set(must) <= set(nat)
 
 
+def _brute_best_score(topo, bdf_of, ids, must, size):
+    """Exhaustive C(n,k) oracle: max pairwise score over every subset of
+    `ids` of cardinality `size` that contains `must`."""
+    import itertools
+    best = -1
+    must_s = set(must)
+    for comb in itertools.combinations(ids, size):
+        if not must_s <= set(comb):
+            continue
+        s = score_set(topo, [bdf_of[d] for d in comb])
+        if s > best:
+            best = s
+    return best
+
+
+@given(topology_instances(), st.data())
+@settings(max_examples=400, deadline=None)
+def test_selector_optimal_vs_bruteforce_oracle(inst, data):
+    """PARITY claim 'brute-force-verified optimal', made true (VERDICT r1
+    item 1): both the Python branch-and-bound and the native C++ selector
+    must attain the exhaustive-enumeration maximum of score_set over all
+    C(n,k) subsets, with and without forced members."""
+    topo, bdf_of = inst
+    ids = sorted(bdf_of)
+    size = data.draw(st.integers(min_value=1, max_value=len(ids)))
+    must = data.draw(st.lists(st.sampled_from(ids),
+                              max_size=min(3, size), unique=True))
+    oracle = _brute_best_score(topo, bdf_of, ids, must, size)
+    for use_native in (False, True):
+        pick = preferred_sets(topo, bdf_of, ids, must, size,
+                              use_native=use_native)
+        assert len(pick) == size, (use_native, pick)
+        assert len(set(pick)) == size
+        assert set(must) <= set(pick)
+        got = score_set(topo, [bdf_of[d] for d in pick])
+        assert got == oracle, (
+            f"native={use_native}: selector score {got} < oracle {oracle} "
+            f"(pick={pick})")
+
+
+@given(topology_instances(), st.data())
+@settings(max_examples=150, deadline=None)
+def test_selector_degraded_budget_path(inst, data):
+    """With the search budget clamped to 1 node the selector must still
+    return a FEASIBLE set (first greedy descent always completes); with the
+    production budget it is optimal (covered by the oracle test)."""
+    from kata_xpu_device_plugin_amd.topology import hive as hive_mod
+    topo, bdf_of = inst
+    ids = sorted(bdf_of)
+    size = data.draw(st.integers(min_value=1, max_value=len(ids)))
+    must = data.draw(st.lists(st.sampled_from(ids),
+                              max_size=min(2, size), unique=True))
+    saved = hive_mod.MAX_SEARCH_NODES
+    hive_mod.MAX_SEARCH_NODES = 1
+    try:
+        pick = preferred_sets(topo, bdf_of, ids, must, size, use_native=False)
+    finally:
+        hive_mod.MAX_SEARCH_NODES = saved
+    assert len(pick) == size
+    assert len(set(pick)) == size
+    assert set(must) <= set(pick) <= set(ids)
+
+
 @given(topology_instances(), st.data())
 @settings(max_examples=100, deadline=None)
 def test_selector_monotone_in_must(inst, data):
