@@ -25,7 +25,6 @@ from __future__ import annotations
 
 import asyncio
 import os
-import queue
 import threading
 import time
 from typing import Dict, List, Optional
@@ -132,32 +131,41 @@ class XPUDevicePlugin:
             devs.append(d)
         return devs
 
-    def _wait_update(self, q: queue.Queue) -> bool:
-        """Blocking wait (run in an executor thread): True once ≥1 update
-        arrived, coalescing bursts into one push."""
-        try:
-            q.get(timeout=0.5)
-        except queue.Empty:
-            return False
-        while True:
-            try:
-                q.get_nowait()
-            except queue.Empty:
-                return True
-
     async def ListAndWatch(self, request, context):
         """Initial full list, then a push per health/inventory change
-        (reference: generic_device_plugin.go:222-250)."""
-        q = self.state.watch()
+        (reference: generic_device_plugin.go:222-250).
+
+        Event-driven bridge: DeviceState fires a thread-safe
+        ``call_soon_threadsafe(event.set)`` per effective change, the
+        stream awaits the asyncio.Event natively — no executor threads, no
+        polling, and no cap on concurrent streams (round 1 served these
+        waits from a 4-thread pool: a 5th stream starved; measured-for in
+        bench.py --clients)."""
         loop = asyncio.get_running_loop()
+        event = asyncio.Event()
+
+        def _notify() -> None:
+            try:
+                loop.call_soon_threadsafe(event.set)
+            except RuntimeError:
+                pass  # loop already closed (shutdown)
+
+        unsubscribe = self.state.subscribe(_notify)
         try:
             yield api.ListAndWatchResponse(devices=self._device_list())
             while not self._stop.is_set():
-                updated = await loop.run_in_executor(None, self._wait_update, q)
-                if updated:
-                    yield api.ListAndWatchResponse(devices=self._device_list())
+                # Bursts coalesce: any number of set() before wait() resumes
+                # produce ONE push of the newest snapshot; a change landing
+                # between clear() and snapshot is included in this push and
+                # re-sets the event → at most one redundant (still correct)
+                # extra push.
+                await event.wait()
+                event.clear()
+                if self._stop.is_set():
+                    break
+                yield api.ListAndWatchResponse(devices=self._device_list())
         finally:
-            self.state.unwatch(q)
+            unsubscribe()
 
     # -- Allocate ------------------------------------------------------
     def _revalidate_many(self, ids: List[str]) -> None:
@@ -345,19 +353,11 @@ class XPUDevicePlugin:
                          name=f"kxdp-register-retry-{self.socket_name}").start()
 
     def _serve_thread(self) -> None:
-        from concurrent.futures import ThreadPoolExecutor
-
         loop = asyncio.new_event_loop()
         self._loop = loop
         asyncio.set_event_loop(loop)
-        # Dedicated bounded executor for the blocking health-queue waits of
-        # ListAndWatch streams (kubelet holds 1-2; don't share the process
-        # default executor with other plugins/components). Shut down
-        # explicitly below: loop.close() does NOT stop executor threads,
-        # which would leak 4 threads per restart cycle.
-        lw_executor = ThreadPoolExecutor(
-            max_workers=4, thread_name_prefix=f"kxdp-lw-{self.socket_name}")
-        loop.set_default_executor(lw_executor)
+        # ListAndWatch is fully event-driven (asyncio bridge, no executor),
+        # so the loop thread is the only thread this server needs.
 
         async def _main():
             server = grpc.aio.server(
@@ -398,7 +398,6 @@ class XPUDevicePlugin:
             self._start_error = e
             self._loop_ready.set()
         finally:
-            lw_executor.shutdown(wait=False, cancel_futures=True)
             try:
                 loop.close()
             except Exception:
@@ -432,6 +431,9 @@ class XPUDevicePlugin:
     def _stop_locked(self) -> None:
         self._stop.set()
         self._serving.clear()
+        # Wake event-driven ListAndWatch streams so they observe _stop and
+        # finish cleanly instead of being cancelled at the grace deadline.
+        self.state.poke()
         # grpc core unlinks the unix socket file itself during server
         # shutdown — mark the removal as self-inflicted BEFORE stopping so
         # the health watcher doesn't treat it as a kubelet wipe.

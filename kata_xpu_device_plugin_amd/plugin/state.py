@@ -10,7 +10,7 @@ from __future__ import annotations
 
 import queue
 import threading
-from typing import Dict, List, Optional, Tuple
+from typing import Callable, Dict, List, Optional, Tuple
 
 from ..discovery.sysfs import XPUDevice
 from ..utils.log import get_logger
@@ -30,6 +30,7 @@ class DeviceState:
         # /dev/vfio node and vice versa.
         self._unhealthy_by: Dict[str, set] = {gid: set() for gid in devices}
         self._watchers: List[queue.Queue] = []
+        self._callbacks: List[Callable[[], None]] = []
         self._generation = 0
 
     # -- snapshots ------------------------------------------------------
@@ -93,6 +94,11 @@ class DeviceState:
                 q.put_nowait(self._generation)
             except queue.Full:
                 pass  # watcher will resync from snapshot anyway
+        for cb in self._callbacks:
+            try:
+                cb()
+            except Exception:
+                pass  # a dead subscriber must not break state mutation
 
     # -- watch ----------------------------------------------------------
     def watch(self) -> queue.Queue:
@@ -107,6 +113,30 @@ class DeviceState:
                 self._watchers.remove(q)
             except ValueError:
                 pass
+
+    def subscribe(self, cb: Callable[[], None]) -> Callable[[], None]:
+        """Register a zero-arg notification callback fired after every
+        effective change (and on poke()). The callback runs under the state
+        lock from the mutating thread: it MUST be non-blocking — the
+        intended use is ``loop.call_soon_threadsafe(event.set)`` from the
+        asyncio ListAndWatch bridge. Returns an idempotent unsubscriber."""
+        with self._lock:
+            self._callbacks.append(cb)
+
+        def unsubscribe() -> None:
+            with self._lock:
+                try:
+                    self._callbacks.remove(cb)
+                except ValueError:
+                    pass
+        return unsubscribe
+
+    def poke(self) -> None:
+        """Wake all watchers without a state change (shutdown path: lets
+        event-driven streams observe their stop flag immediately instead of
+        waiting out the gRPC shutdown grace)."""
+        with self._lock:
+            self._notify_locked()
 
 
 def _gkey(g: str):
