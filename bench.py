@@ -120,6 +120,14 @@ def main() -> int:
     grpc.channel_ready_future(channel).result(timeout=10)
     plugin = api.DevicePluginStub(channel)
 
+    # Client-side GC hygiene mirroring the daemon's (config.gc_tuning):
+    # measured latency includes the CLIENT's own interpreter pauses, and a
+    # generational collection mid-RPC shows up as a fake multi-ms tail.
+    import gc
+    gc.collect()
+    gc.freeze()
+    gc.set_threshold(50_000, 20, 20)
+
     all_ids = [str(70 + i) for i in range(args.node_gpus)]
     gpus_per_pod = max(1, min(args.gpus, args.node_gpus))
 

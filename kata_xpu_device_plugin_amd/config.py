@@ -134,6 +134,11 @@ class Config:
     # 0 = off. Rescans that find no change are no-ops (no CDI rewrite).
     rescan_interval_s: float = field(default_factory=lambda: _env_float("RESCAN_S", 0.0))
 
+    # --- performance ---
+    # Freeze the startup object graph + raise GC thresholds after start()
+    # (kills multi-ms generational-GC pauses in Allocate's p99 tail).
+    gc_tuning: bool = field(default_factory=lambda: _env_bool("GC_TUNING", True))
+
     # --- observability ---
     metrics_port: int = field(default_factory=lambda: _env_int("METRICS_PORT", 0))  # 0 = off
     log_level: str = field(default_factory=lambda: _env("LOG_LEVEL", "INFO"))

@@ -101,6 +101,7 @@ class XPUDevicePlugin:
         self._devices_dir = os.path.join(cfg.sysfs_root, "bus", "pci", "devices")
         self._vendor_list = list(cfg.vendor_allowlist)
         self._qn_cache: Dict[str, str] = {}      # gid → "kind=gid"
+        self._bdf_cache = None                   # (generation, gid→bdf)
         self._env_res_name = ENV_PCI_RESOURCE_PREFIX + resource_name.upper(
         ).replace("/", "_").replace(".", "_").replace("-", "_")
 
@@ -265,11 +266,16 @@ class XPUDevicePlugin:
 
     async def GetPreferredAllocation(self, request, context):
         resp = api.PreferredAllocationResponse()
-        bdf_of: Dict[str, str] = {}
-        for gid in self.state.device_ids():
-            dev = self.state.device(gid)
-            if dev:
-                bdf_of[gid] = dev.primary.bdf
+        # gid→BDF map cached per state generation: rebuilt only when the
+        # device set or health actually changed, not per admission RPC
+        # (shaves loop occupancy → queueing tail under concurrent clients).
+        gen = self.state.generation
+        cached = self._bdf_cache
+        if cached is not None and cached[0] == gen:
+            bdf_of = cached[1]
+        else:
+            bdf_of = self.state.bdf_map()
+            self._bdf_cache = (gen, bdf_of)
         for creq in request.container_requests:
             available = [d for d in creq.available_device_ids if d in bdf_of]
             must = list(creq.must_include_device_ids)

@@ -57,6 +57,18 @@ class DeviceState:
             u = self._unhealthy_by.get(gid)
             return u is not None and not u
 
+    @property
+    def generation(self) -> int:
+        """Monotonic change counter (device set or effective health);
+        lets hot paths cache derived maps keyed on it."""
+        with self._lock:
+            return self._generation
+
+    def bdf_map(self) -> Dict[str, str]:
+        """gid → primary BDF for the current device set (one lock trip)."""
+        with self._lock:
+            return {gid: dev.primary.bdf for gid, dev in self._devices.items()}
+
     # -- mutation -------------------------------------------------------
     def set_health(self, gid: str, healthy: bool, source: str = "vfio") -> bool:
         """Record one source's verdict; returns True if the device's
