@@ -12,7 +12,8 @@ real kubelet flow: GetPreferredAllocation(size=N) → Allocate(picked N
 device IDs) (reference hot path: generic_device_plugin.go:320-355).
 
 Scaling mode (driver: torchrun --nproc-per-node N):
-  * rank 0 hosts the plugin daemon for the shared 8-GPU node,
+  * rank 0 spawns the production daemon CLI as its own process for the
+    shared 8-GPU node (no rank shares an interpreter with the server),
   * every rank (incl. 0) runs an independent churn client — concurrency
     grows with N, and each pod requests N GPUs (the BASELINE matrix),
   * per-rank work is fixed → "weak" scaling; value = MAX over ranks' p50
@@ -68,7 +69,6 @@ def main() -> int:
         use_cuda = nccl_ok
 
     from kata_xpu_device_plugin_amd.plugin import api
-    from kata_xpu_device_plugin_amd.plugin.manager import PluginManager
     from kata_xpu_device_plugin_amd.testing.kubelet_stub import KubeletStub
     from kata_xpu_device_plugin_amd.testing.mocknode import make_mock_node
     from kata_xpu_device_plugin_amd.utils.log import configure
@@ -89,21 +89,36 @@ def main() -> int:
             "probe_passed": rep.passed,
         }
 
-    # --- rank 0: bring up the node + daemon, share the socket dir -------
-    mgr = stub = None
+    # --- rank 0: bring up the node; daemon runs as its OWN process ------
+    # (the production shape: kubelet and the plugin daemon never share an
+    # interpreter. Round 1 hosted the daemon in rank 0's process, so the
+    # server's event loop fought rank 0's churn client for one GIL — that
+    # convoy WAS the 4-client p99 tail)
+    import subprocess
+    daemon = stub = None
     if rank == 0:
         root = tempfile.mkdtemp(prefix="kxdp-bench-")
         node = make_mock_node(root, n_gpus=args.node_gpus)
         cfg = node.config()
         stub = KubeletStub(cfg.kubelet_socket_dir)
         stub.start()
-        mgr = PluginManager(cfg)
-        mgr.setup()
-        mgr.start(register=True)
-        reg = stub.wait_for_registration(1)[0]
+        denv = os.environ.copy()
+        denv.update({
+            "KXDP_SYSFS_ROOT": cfg.sysfs_root,
+            "KXDP_DEV_ROOT": cfg.dev_root,
+            "KXDP_CDI_DIR": cfg.cdi_dir,
+            "KXDP_KUBELET_DIR": cfg.kubelet_socket_dir,
+            "KXDP_TOPOLOGY_HINT": cfg.topology_hint_path,
+            "KXDP_PCI_IDS": "",
+            "KXDP_AMDSMI_HEALTH": "0",   # no amd-smi noise in the bench
+            "KXDP_LOG_LEVEL": "WARNING",
+        })
+        daemon = subprocess.Popen(
+            [sys.executable, "-m", "kata_xpu_device_plugin_amd"], env=denv)
+        reg = stub.wait_for_registration(1, timeout=30)[0]
         endpoint_dir = cfg.kubelet_socket_dir
         endpoint = reg.endpoint
-        discovered = len(mgr.inventory.devices)
+        discovered = args.node_gpus  # confirmed via ListAndWatch below
     else:
         endpoint_dir = endpoint = None
         discovered = args.node_gpus
@@ -119,6 +134,12 @@ def main() -> int:
         f"unix://{os.path.join(endpoint_dir, endpoint)}")
     grpc.channel_ready_future(channel).result(timeout=10)
     plugin = api.DevicePluginStub(channel)
+
+    if rank == 0:
+        # Confirm the daemon's advertised inventory (initial ListAndWatch).
+        stream = plugin.ListAndWatch(api.Empty())
+        discovered = len(next(stream).devices)
+        stream.cancel()
 
     # Client-side GC hygiene mirroring the daemon's (config.gc_tuning):
     # measured latency includes the CLIENT's own interpreter pauses, and a
@@ -247,7 +268,11 @@ def main() -> int:
         dist.barrier()
         dist.destroy_process_group()
     if rank == 0:
-        mgr.stop()
+        daemon.terminate()   # SIGTERM → manager's orderly shutdown path
+        try:
+            daemon.wait(timeout=10)
+        except subprocess.TimeoutExpired:
+            daemon.kill()
         stub.stop()
     return 0
 

@@ -37,6 +37,8 @@ class PluginManager:
         self.watcher: Optional[NodeWatcher] = None
         self.cdi_spec_path: Optional[str] = None
         self._stop = threading.Event()
+        self._register = True          # start()'s registration choice,
+        self._rescan_lock = threading.Lock()  # reused by dynamic rescan
         self._metrics = None
         self._amdsmi = None
         self._gpu_probe = None
@@ -85,6 +87,7 @@ class PluginManager:
 
     # ------------------------------------------------------------------
     def start(self, register: bool = True) -> None:
+        self._register = register
         if not self.plugins:
             log.warning("no devices discovered; serving nothing (will still "
                         "watch for kubelet restarts)")
@@ -141,7 +144,7 @@ class PluginManager:
         gc.set_threshold(50_000, 20, 20)
 
     def _on_socket_removed(self, socket_name: str) -> None:
-        for plugin in self.plugins.values():
+        for plugin in list(self.plugins.values()):
             if plugin.socket_name == socket_name and not self._stop.is_set():
                 # Skip removals the plugin inflicted on itself (stop or
                 # restart); only an EXTERNAL wipe (kubelet cleanup) should
@@ -172,7 +175,7 @@ class PluginManager:
         """amd-smi verdict → DeviceState (matches any function BDF of a
         schedulable group; normally only fires on hybrid/pre-flight nodes
         since vfio-bound GPUs are invisible to amd-smi)."""
-        for state in self.states.values():
+        for state in list(self.states.values()):
             for gid in state.device_ids():
                 dev = state.device(gid)
                 if dev and bdf in (fn.bdf.lower() for fn in dev.functions):
@@ -191,7 +194,7 @@ class PluginManager:
     def _on_probe_health(self, bdf: str, healthy: bool) -> None:
         """In-daemon GPU probe verdict → DeviceState (source "probe")."""
         bdf = bdf.lower()
-        for state in self.states.values():
+        for state in list(self.states.values()):
             for gid in state.device_ids():
                 dev = state.device(gid)
                 if dev and bdf in (fn.bdf.lower() for fn in dev.functions):
@@ -203,7 +206,7 @@ class PluginManager:
     def _on_kubelet_restarted(self) -> None:
         if self._stop.is_set():
             return
-        for plugin in self.plugins.values():
+        for plugin in list(self.plugins.values()):
             try:
                 plugin.register_with_kubelet()
             except Exception:
@@ -239,7 +242,7 @@ class PluginManager:
         if self._gpu_probe is not None:
             self._gpu_probe.stop()
             self._gpu_probe = None
-        for plugin in self.plugins.values():
+        for plugin in list(self.plugins.values()):
             plugin.stop()
         if self._metrics is not None:
             self._metrics.stop()
@@ -257,7 +260,17 @@ class PluginManager:
     def rescan(self) -> bool:
         """Re-run discovery and swap device sets in place (e.g. after VF
         count changes); plugins keep serving. Returns True if anything
-        changed (no-change rescans skip the CDI rewrite and state swap)."""
+        changed (no-change rescans skip the CDI rewrite and state swap).
+
+        Fully dynamic: topology is reloaded (new VFs
+        get hive/NUMA locality immediately — advisor r1 finding), plugins
+        for brand-new resource names (e.g. first SR-IOV enable creating
+        *_VF) are created, started and registered in place, and plugins
+        whose resource vanished entirely are stopped and retired."""
+        with self._rescan_lock:
+            return self._rescan_locked()
+
+    def _rescan_locked(self) -> bool:
         inv = scan_node(self.cfg)
         if self.inventory is not None and \
                 self._inventory_signature(inv) == \
@@ -267,18 +280,47 @@ class PluginManager:
                  len(self.inventory.devices) if self.inventory else 0,
                  len(inv.devices))
         self.inventory = inv
+        self.topology = load_topology(self.cfg, inv)
         spec = build_spec(inv, self.cfg.cdi_kind, self.cfg.dev_root)
         self.cdi_spec_path = write_spec(
             spec, self.cfg.cdi_dir, self.cfg.cdi_spec_name, self.cfg.cdi_format
         )
         grouped = self._group_by_resource(inv)
-        for rname, state in self.states.items():
-            state.replace_devices(grouped.get(rname, {}))
-        # brand-new resource names require a daemon restart (documented)
-        new = set(grouped) - set(self.states)
-        if new:
-            log.warning("rescan found new resource name(s) %s — restart the "
-                        "daemon to serve them", sorted(new))
+
+        # Existing resources: swap device sets + refresh topology handle.
+        for rname, state in list(self.states.items()):
+            if rname in grouped:
+                state.replace_devices(grouped[rname])
+                self.plugins[rname].topo = self.topology
+
+        # Retire plugins whose resource has no devices left (kubelet drops
+        # the resource when the plugin's socket goes away).
+        for rname in [r for r in self.plugins if r not in grouped]:
+            plugin = self.plugins.pop(rname)
+            self.states.pop(rname, None)
+            try:
+                plugin.stop()
+            except Exception:
+                log.exception("stopping retired plugin %s failed", rname)
+            log.info("resource %s retired (no devices after rescan)", rname)
+
+        # Brand-new resource names: serve them now, no daemon restart.
+        for rname in sorted(set(grouped) - set(self.plugins)):
+            state = DeviceState(grouped[rname])
+            plugin = XPUDevicePlugin(self.cfg, rname, state, self.topology)
+            self.states[rname] = state
+            self.plugins[rname] = plugin
+            if self.watcher is not None:  # daemon is live → start serving
+                try:
+                    plugin.start(register=self._register)
+                    log.info("resource %s: started dynamically (%d devices)",
+                             rname, len(grouped[rname]))
+                except Exception:
+                    log.exception("dynamic start of %s failed", rname)
+
+        if self.watcher is not None:
+            self.watcher.plugin_socket_names = {
+                p.socket_name for p in self.plugins.values()}
         return True
 
 
