@@ -142,6 +142,28 @@ class MockNode:
                 )
         self.gpus.append(gpu)
 
+    def remove_gpu(self, bdf: str) -> None:
+        """Inverse of add_gpu: delete the PCI function (and its IOMMU-group
+        membership + vfio node when it was the last member) — models VF
+        disable / device unplug for rescan tests."""
+        import shutil
+        gpu = next((g for g in self.gpus if g.bdf == bdf), None)
+        d = self._pci_dir(bdf)
+        if os.path.isdir(d):
+            shutil.rmtree(d)
+        if gpu is not None:
+            self.gpus.remove(gpu)
+            if gpu.iommu_group:
+                gdir = os.path.join(self.sysfs, "kernel", "iommu_groups",
+                                    gpu.iommu_group)
+                member = os.path.join(gdir, "devices", bdf)
+                if os.path.exists(member):
+                    os.unlink(member)
+                remaining = os.listdir(os.path.join(gdir, "devices")) \
+                    if os.path.isdir(os.path.join(gdir, "devices")) else []
+                if not remaining:
+                    self.remove_vfio_node(gpu.iommu_group)
+
     # --- /dev/vfio ----------------------------------------------------
     def add_vfio_node(self, group: str) -> str:
         path = os.path.join(self.dev, "vfio", group)

@@ -246,6 +246,69 @@ def test_mixed_pf_vf_node(tmp_path):
         stub.stop()
 
 
+def test_rescan_serves_new_resource_name_without_restart(tmp_path):
+    """First-time SR-IOV enable creates a brand-new *_VF resource name:
+    rescan() must create, start and register its plugin in place — and
+    retire it again when the VFs disappear (round 1
+    logged 'restart the daemon' instead)."""
+    node = make_mock_node(str(tmp_path), n_gpus=2, kfd=False, hint=True)
+    cfg = node.config()
+    stub = KubeletStub(cfg.kubelet_socket_dir)
+    stub.start()
+    mgr = PluginManager(cfg)
+    mgr.setup()
+    mgr.start(register=True)
+    try:
+        stub.wait_for_registration(1)
+        assert sorted(mgr.plugins) == ["amd.com/INSTINCT_MI355X"]
+
+        # VFs appear (e.g. tools/sriov enable): new resource name.
+        for k in range(2):
+            node.add_gpu(MockGPU(bdf=f"0000:61:02.{k}", device_id=0x75B3,
+                                 iommu_group=str(120 + k),
+                                 physfn_bdf="0000:61:00.0"))
+        assert mgr.rescan() is True
+        regs = stub.wait_for_registration(2)
+        names = sorted(r.resource_name for r in regs)
+        assert names == ["amd.com/INSTINCT_MI355X", "amd.com/INSTINCT_MI355X_VF"]
+        assert sorted(mgr.plugins) == names
+
+        # The dynamic plugin serves for real: Allocate a VF end-to-end.
+        vf_reg = next(r for r in regs if r.resource_name.endswith("_VF"))
+        ps = stub.plugin_stub(vf_reg.endpoint)
+        resp = ps.Allocate(api.AllocateRequest(container_requests=[
+            api.ContainerAllocateRequest(devices_ids=["120"])]))
+        assert resp.container_responses[0].cdi_devices[0].name == "amd.com/gpu=120"
+
+        # Topology was reloaded (advisor r1): the VFs have locality — both
+        # share their PF's (pseudo-)hive, so a 2-VF pod scores as one hive.
+        assert mgr.topology.hive_of.get("0000:61:02.0") is not None
+        assert mgr.topology.hive_of["0000:61:02.0"] == \
+            mgr.topology.hive_of["0000:61:02.1"]
+        pref = ps.GetPreferredAllocation(api.PreferredAllocationRequest(
+            container_requests=[api.ContainerPreferredAllocationRequest(
+                available_device_ids=["120", "121"], allocation_size=2)]))
+        assert sorted(pref.container_responses[0].device_ids) == ["120", "121"]
+
+        # VFs disappear again → the *_VF plugin is retired.
+        vf_socket = mgr.plugins["amd.com/INSTINCT_MI355X_VF"].socket_path
+        for k in range(2):
+            node.remove_gpu(f"0000:61:02.{k}")
+        assert mgr.rescan() is True
+        assert sorted(mgr.plugins) == ["amd.com/INSTINCT_MI355X"]
+        assert not os.path.exists(vf_socket)
+        # PF resource keeps serving untouched
+        pf_reg = next(r for r in regs if not r.resource_name.endswith("_VF"))
+        ps_pf = stub.plugin_stub(pf_reg.endpoint)
+        assert len(ps_pf.GetPreferredAllocation(api.PreferredAllocationRequest(
+            container_requests=[api.ContainerPreferredAllocationRequest(
+                available_device_ids=["70", "71"], allocation_size=1)]
+        )).container_responses[0].device_ids) == 1
+    finally:
+        mgr.stop()
+        stub.stop()
+
+
 def test_daemon_survives_missing_kubelet(tmp_path):
     """kubelet not up at daemon start (node-boot race): the plugin must
     serve anyway and register as soon as kubelet appears."""
