@@ -369,6 +369,11 @@ class XPUDevicePlugin:
             server = grpc.aio.server(
                 options=[("grpc.max_concurrent_streams", 64)])
             api.add_device_plugin_servicer(server, self)
+            # The socket dir normally exists (kubelet / hostPath
+            # DirectoryOrCreate); create it ourselves so a daemon started
+            # before kubelet — or restarted after a kubelet-dir wipe — can
+            # still bind and serve.
+            os.makedirs(self.cfg.kubelet_socket_dir, exist_ok=True)
             if os.path.exists(self.socket_path):
                 self._note_expected_removal()
                 os.unlink(self.socket_path)

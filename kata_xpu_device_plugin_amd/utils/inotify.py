@@ -24,6 +24,8 @@ IN_MOVED_TO = 0x00000080
 IN_DELETE_SELF = 0x00000400
 IN_ATTRIB = 0x00000004
 IN_ONLYDIR = 0x01000000
+IN_IGNORED = 0x00008000   # kernel dropped the watch (dir deleted)
+IN_MOVE_SELF = 0x00000800  # watched dir renamed away (watch follows inode!)
 
 _EVENT_HDR = struct.Struct("iIII")  # wd, mask, cookie, len
 
@@ -43,6 +45,13 @@ class Event:
     @property
     def removed(self) -> bool:
         return bool(self.mask & (IN_DELETE | IN_MOVED_FROM | IN_DELETE_SELF))
+
+    @property
+    def ignored(self) -> bool:
+        """Watch no longer observes the original path: the dir was deleted
+        (IN_IGNORED) or renamed away (IN_MOVE_SELF — inotify watches follow
+        the inode, so events would silently track the MOVED dir)."""
+        return bool(self.mask & (IN_IGNORED | IN_MOVE_SELF))
 
 
 class Inotify:

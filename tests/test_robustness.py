@@ -309,6 +309,114 @@ def test_rescan_serves_new_resource_name_without_restart(tmp_path):
         stub.stop()
 
 
+def _wait_until(pred, timeout=5.0, what="condition"):
+    deadline = time.monotonic() + timeout
+    while time.monotonic() < deadline:
+        if pred():
+            return
+        time.sleep(0.05)
+    raise AssertionError(f"timed out waiting for {what}")
+
+
+def test_watcher_kubelet_dir_created_after_start(tmp_path):
+    """a kubelet dir created AFTER the watcher started
+    must still be watched — a subsequent socket wipe triggers the restart
+    callback, and kubelet.sock creation triggers re-registration. Round 1
+    armed the watch only if the dir existed at startup."""
+    from kata_xpu_device_plugin_amd.config import Config
+    from kata_xpu_device_plugin_amd.health.watcher import NodeWatcher
+
+    kdir = os.path.join(str(tmp_path), "var", "lib", "kubelet",
+                        "device-plugins")
+    cfg = Config(
+        sysfs_root=os.path.join(str(tmp_path), "sys"),
+        dev_root=os.path.join(str(tmp_path), "dev"),
+        cdi_dir=os.path.join(str(tmp_path), "cdi"),
+        kubelet_socket_dir=kdir,
+        pci_ids_paths=(),
+    )
+    removed, restarted = [], []
+    w = NodeWatcher(
+        cfg, {},
+        on_socket_removed=removed.append,
+        on_kubelet_restarted=lambda: restarted.append(1),
+        plugin_socket_names={"kxdp-test.sock"},
+    )
+    w.start()
+    w.wait_ready()
+    try:
+        assert not os.path.isdir(kdir)
+        os.makedirs(kdir)                      # kubelet dir appears late
+        _wait_until(lambda: "kubelet" in w._armed, what="late dir armed")
+        # (arming a late dir may fire one reconcile restart for the
+        # not-yet-recreated socket — correct behavior; count from here)
+        sock = os.path.join(kdir, "kxdp-test.sock")
+        open(sock, "w").close()
+        time.sleep(0.3)
+        base = len(removed)
+        os.unlink(sock)                        # the wipe we must catch
+        _wait_until(lambda: len(removed) > base and
+                    removed[-1] == "kxdp-test.sock",
+                    what="socket-wipe callback")
+        open(os.path.join(kdir, "kubelet.sock"), "w").close()
+        _wait_until(lambda: restarted, what="kubelet-restart callback")
+    finally:
+        w.stop()
+
+
+def test_kubelet_dir_wipe_full_recovery(tmp_path):
+    """The whole kubelet dir vanishes mid-run (kubelet reinstall; modeled
+    as an atomic rename — inotify watches follow the inode, so round 1's
+    watcher silently kept watching the MOVED dir): the plugin restarts,
+    recreates dir + socket, and the RE-ARMED watch still catches a later
+    plain socket wipe."""
+    node = make_mock_node(str(tmp_path), n_gpus=1)
+    cfg = node.config()
+    stub = KubeletStub(cfg.kubelet_socket_dir)
+    stub.start()
+    mgr = PluginManager(cfg)
+    mgr.setup()
+    mgr.start(register=True)
+    plugin = next(iter(mgr.plugins.values()))
+    try:
+        stub.wait_for_registration(1)
+        stub.stop()                       # free kubelet.sock before the wipe
+        os.rename(cfg.kubelet_socket_dir, cfg.kubelet_socket_dir + ".old")
+        _wait_until(lambda: plugin.serving and
+                    os.path.exists(plugin.socket_path), timeout=10,
+                    what="socket recreation after dir wipe")
+        time.sleep(0.8)                   # watcher re-armed on the new dir
+        os.unlink(plugin.socket_path)     # later plain wipe must be caught
+        _wait_until(lambda: plugin.serving and
+                    os.path.exists(plugin.socket_path), timeout=10,
+                    what="restart after post-wipe socket removal")
+    finally:
+        mgr.stop()
+
+
+def test_cdi_dir_wipe_selfheal(tmp_path):
+    """The entire CDI dir vanishes (tmp-cleaner; atomic rename so the wipe
+    cannot race the self-heal): dir AND spec are regenerated. Round 1 only
+    healed a removed spec FILE inside a surviving dir."""
+    import shutil
+    node = make_mock_node(str(tmp_path), n_gpus=2)
+    cfg = node.config()
+    mgr = PluginManager(cfg)
+    mgr.setup()
+    mgr.start(register=False)
+    try:
+        path = mgr.cdi_spec_path
+        assert path and os.path.exists(path)
+        os.rename(cfg.cdi_dir, cfg.cdi_dir + ".old")
+        shutil.rmtree(cfg.cdi_dir + ".old")
+        _wait_until(lambda: os.path.exists(path), timeout=6.0,
+                    what="CDI spec regeneration after dir wipe")
+        spec = read_spec(path)
+        assert spec.device_names() == ["70", "71"]
+    finally:
+        mgr.stop()
+
+
 def test_daemon_survives_missing_kubelet(tmp_path):
     """kubelet not up at daemon start (node-boot race): the plugin must
     serve anyway and register as soon as kubelet appears."""
