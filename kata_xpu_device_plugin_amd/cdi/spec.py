@@ -148,6 +148,12 @@ def write_spec(spec: CDISpec, cdi_dir: str, spec_name: str, fmt: str = "yaml") -
     os.makedirs(cdi_dir, exist_ok=True)
     path = spec_path(cdi_dir, spec_name, fmt)
     obj = spec.to_obj()
+    # Hard contract check: nothing schema-invalid ever reaches /var/run/cdi
+    # (the runtime would reject or — worse — misparse it).
+    from .schema import validate_spec_obj
+    problems = validate_spec_obj(obj)
+    if problems:
+        raise ValueError(f"CDI spec fails schema validation: {problems}")
     if fmt == "yaml":
         payload = yaml.safe_dump(obj, sort_keys=False)
     else:

@@ -58,6 +58,26 @@ class PluginManager:
             out.setdefault(rname, {})[gid] = dev
         return out
 
+    def _write_cdi(self, inv: NodeInventory) -> Optional[str]:
+        """Write the CDI spec for this inventory; a zero-device spec is
+        schema-invalid (upstream CDI rejects empty `devices`), so empty
+        nodes get no spec file — and any stale one is removed."""
+        if not inv.devices:
+            from ..cdi.spec import spec_path
+            for fmt in ("yaml", "json"):
+                stale = spec_path(self.cfg.cdi_dir, self.cfg.cdi_spec_name, fmt)
+                if os.path.exists(stale):
+                    try:
+                        os.unlink(stale)
+                    except OSError:
+                        pass
+            log.info("no devices: no CDI spec written")
+            return None
+        spec = build_spec(inv, self.cfg.cdi_kind, self.cfg.dev_root)
+        return write_spec(
+            spec, self.cfg.cdi_dir, self.cfg.cdi_spec_name, self.cfg.cdi_format
+        )
+
     def setup(self) -> None:
         """Discovery pass → CDI spec → per-resource plugin objects."""
         inv = scan_node(self.cfg)
@@ -71,10 +91,7 @@ class PluginManager:
         log.info("xGMI topology source: %s (%d hive-mapped GPUs)",
                  self.topology.source, len(self.topology.hive_of))
 
-        spec = build_spec(inv, self.cfg.cdi_kind, self.cfg.dev_root)
-        self.cdi_spec_path = write_spec(
-            spec, self.cfg.cdi_dir, self.cfg.cdi_spec_name, self.cfg.cdi_format
-        )
+        self.cdi_spec_path = self._write_cdi(inv)
 
         for rname, devs in sorted(self._group_by_resource(inv).items()):
             state = DeviceState(devs)
@@ -164,10 +181,7 @@ class PluginManager:
         if self._stop.is_set() or self.inventory is None:
             return
         try:
-            spec = build_spec(self.inventory, self.cfg.cdi_kind, self.cfg.dev_root)
-            self.cdi_spec_path = write_spec(
-                spec, self.cfg.cdi_dir, self.cfg.cdi_spec_name, self.cfg.cdi_format
-            )
+            self.cdi_spec_path = self._write_cdi(self.inventory)
         except Exception:
             log.exception("CDI spec regeneration failed")
 
@@ -281,10 +295,7 @@ class PluginManager:
                  len(inv.devices))
         self.inventory = inv
         self.topology = load_topology(self.cfg, inv)
-        spec = build_spec(inv, self.cfg.cdi_kind, self.cfg.dev_root)
-        self.cdi_spec_path = write_spec(
-            spec, self.cfg.cdi_dir, self.cfg.cdi_spec_name, self.cfg.cdi_format
-        )
+        self.cdi_spec_path = self._write_cdi(inv)
         grouped = self._group_by_resource(inv)
 
         # Existing resources: swap device sets + refresh topology handle.

@@ -36,7 +36,13 @@ def main(argv=None) -> int:
     cdi_dir = cfg.cdi_dir if os.access(cfg.cdi_dir, os.W_OK) else \
         tempfile.mkdtemp(prefix="kxdp-validate-")
     spec = build_spec(inv, cfg.cdi_kind, cfg.dev_root)
-    write_spec(spec, cdi_dir, cfg.cdi_spec_name, cfg.cdi_format)
+    spec_file = write_spec(spec, cdi_dir, cfg.cdi_spec_name, cfg.cdi_format)
+    from ..cdi.schema import validate_spec_file
+    schema_problems = validate_spec_file(spec_file)
+    if schema_problems:
+        print(f"FAIL CDI schema: {schema_problems}")
+        return 1
+    print(f"CDI spec {spec_file}: schema-valid (CDI {spec.cdi_version})")
     resolver = CDIResolver(cdi_dir)
 
     rc = 0
