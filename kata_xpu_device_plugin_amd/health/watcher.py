@@ -167,6 +167,11 @@ class NodeWatcher(threading.Thread):
                         rearm |= self._handle(ev)
                     except Exception:  # watcher must never die silently
                         log.exception("health watcher event error: %s", ev)
+                # While any target dir is unarmed, retry every tick: a deep
+                # mkdir -p chain outruns event-driven re-arming (each dir
+                # watch only sees DIRECT children, and the next path
+                # component may exist before its ancestor watch is armed).
+                rearm |= len(self._armed) < 3
                 if rearm and not self._stop_evt.is_set():
                     try:
                         self._arm_all(ino)
