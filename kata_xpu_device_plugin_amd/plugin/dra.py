@@ -17,11 +17,14 @@ implements the kubelet-facing half for this driver:
   (`resource_slice_obj()`) for the control-plane publisher (publishing
   to the API server needs a cluster and is out of scope here).
 
-STATUS: experimental. The protobuf schema mirrors
-``k8s.io/kubelet/pkg/apis/dra/v1beta1/api.proto`` (k8s 1.32) from
-documentation; field numbers MUST be re-verified against the target
-kubelet before production use (no protoc/vendored proto available in
-this build environment). The preparation logic and tests are real.
+STATUS: experimental (needs a DRA-enabled cluster for end-to-end
+verification). The protobuf schema mirrors
+``k8s.io/kubelet/pkg/apis/dra/v1beta1/api.proto`` (k8s 1.32) and is
+pinned against the transcription checked into
+``tests/data/dra_v1beta1.proto`` by ``tests/test_proto_pin.py`` —
+every message, field number/type/label and the DRAPlugin service path
+are asserted there (same pinning as the DevicePlugin/pluginregistration/
+podresources schemas). The preparation logic and tests are real.
 """
 from __future__ import annotations
 
