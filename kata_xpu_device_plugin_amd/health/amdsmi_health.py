@@ -29,17 +29,91 @@ _FATAL_TEMP_C = 105.0
 
 
 @dataclass
+class XgmiLinkHealth:
+    index: int
+    status: str = "unknown"      # up | down | disabled | unknown
+    errors: int = 0              # cumulative link error count
+
+
+@dataclass
 class DeviceHealth:
     bdf: str
     healthy: bool = True
     reasons: List[str] = field(default_factory=list)
     temperature_c: Optional[float] = None
     uncorrectable_errors: int = 0
+    # Per-xGMI-link telemetry. A sick link does NOT
+    # make the GPU Unhealthy (it still computes) — it degrades the hive in
+    # placement (topology/hive.py::GPUTopology.degraded) and is exported
+    # per-link to Prometheus.
+    xgmi_links: List[XgmiLinkHealth] = field(default_factory=list)
+
+    @property
+    def xgmi_sick(self) -> bool:
+        return any(l.errors > 0 or l.status == "down" for l in self.xgmi_links)
 
 
 def _normalize_bdf(raw: str) -> str:
     # amdsmi returns "0000:0a:00.0" already; be liberal in what we accept.
     return raw.strip().lower()
+
+
+def _read_xgmi_links(amdsmi, h) -> List[XgmiLinkHealth]:
+    """Best-effort per-link xGMI status/error read. The amdsmi surface for
+    this moved across ROCm releases, so try the dedicated APIs first and
+    fall back to gpu_metrics' xgmi_link_status array; every call is
+    defensive (missing API ⇒ fewer fields, never an exception out)."""
+    links: List[XgmiLinkHealth] = []
+    # 1) per-link status (amdsmi_get_gpu_xgmi_link_status, ROCm >= 6.3)
+    try:
+        st = amdsmi.amdsmi_get_gpu_xgmi_link_status(h)
+        # dict {"status": [..]} or plain list depending on version
+        raw = st.get("status") if isinstance(st, dict) else st
+        for i, v in enumerate(raw or []):
+            name = str(v).rsplit(".", 1)[-1].lower()
+            if name.isdigit():  # numeric enum: 0=disabled,1=up,2=down (smi)
+                name = {0: "disabled", 1: "up", 2: "down"}.get(int(name),
+                                                               "unknown")
+            links.append(XgmiLinkHealth(index=i, status=name))
+    except Exception:
+        pass
+    if not links:
+        # 2) gpu_metrics fallback
+        try:
+            gm = amdsmi.amdsmi_get_gpu_metrics_info(h)
+            raw = gm.get("xgmi_link_status", []) if isinstance(gm, dict) else []
+            for i, v in enumerate(raw):
+                try:
+                    iv = int(v)
+                except (TypeError, ValueError):
+                    continue
+                if iv in (0xFFFF, 0xFFFFFFFF):  # UNSUPPORTED sentinel
+                    continue
+                links.append(XgmiLinkHealth(
+                    index=i, status={0: "down", 1: "up"}.get(iv, "unknown")))
+        except Exception:
+            pass
+    # 3) cumulative error status (per-device; attribute to all links when
+    # the per-link error API is absent)
+    err_total = 0
+    try:
+        es = amdsmi.amdsmi_gpu_xgmi_error_status(h)
+        # enum: 0 NO_ERRORS, 1 ERROR, 2 MULTIPLE_ERRORS
+        name = str(es).rsplit(".", 1)[-1].upper()
+        if name in ("ERROR",):
+            err_total = 1
+        elif name in ("MULTIPLE_ERRORS",):
+            err_total = 2
+        elif name.isdigit():
+            err_total = int(name)
+    except Exception:
+        pass
+    if err_total and not links:
+        links.append(XgmiLinkHealth(index=0, status="unknown",
+                                    errors=err_total))
+    elif err_total:
+        links[0].errors = err_total
+    return links
 
 
 def snapshot() -> Dict[str, DeviceHealth]:
@@ -90,6 +164,7 @@ def snapshot() -> Dict[str, DeviceHealth]:
                     dh.reasons.append(f"junction {dh.temperature_c:.0f}°C")
             except Exception:
                 pass
+            dh.xgmi_links = _read_xgmi_links(amdsmi, h)
             out[dh.bdf] = dh
     finally:
         try:
@@ -108,21 +183,28 @@ class AmdSmiPoller(threading.Thread):
         interval_s: float,
         on_health: Callable[[str, bool, List[str]], None],
         snapshot_fn: Callable[[], Dict[str, DeviceHealth]] = snapshot,
+        on_xgmi: Optional[Callable[[Dict[str, DeviceHealth]], None]] = None,
     ):
         super().__init__(name="kxdp-amdsmi-poller", daemon=True)
         self.interval_s = interval_s
         self.on_health = on_health
+        self.on_xgmi = on_xgmi
         self.snapshot_fn = snapshot_fn
         self._stop_evt = threading.Event()
         self._last: Dict[str, bool] = {}
+        self.last_snapshot: Dict[str, DeviceHealth] = {}  # metrics export
 
     def poll_once(self) -> None:
-        for bdf, dh in self.snapshot_fn().items():
+        snap = self.snapshot_fn()
+        self.last_snapshot = snap
+        for bdf, dh in snap.items():
             prev = self._last.get(bdf)
             if prev is None or prev != dh.healthy:
                 self._last[bdf] = dh.healthy
                 if prev is not None or not dh.healthy:
                     self.on_health(bdf, dh.healthy, dh.reasons)
+        if self.on_xgmi is not None and snap:
+            self.on_xgmi(snap)
 
     def run(self) -> None:
         while not self._stop_evt.wait(self.interval_s):

@@ -44,6 +44,17 @@ class MetricsExporter:
             "kxdp_discovery_seconds", "Wall time of the last discovery scan",
             registry=self.registry,
         )
+        self.g_xgmi_up = Gauge(
+            "kxdp_xgmi_link_up",
+            "xGMI link state (1 up, 0 down/disabled) per GPU link "
+            "(amd-smi; amdgpu-bound GPUs only)",
+            ["bdf", "link"], registry=self.registry,
+        )
+        self.g_hive_degraded = Gauge(
+            "kxdp_xgmi_degraded",
+            "GPU excluded from hive placement due to sick xGMI links",
+            ["bdf"], registry=self.registry,
+        )
 
     def refresh(self) -> None:
         m = self.manager
@@ -54,6 +65,15 @@ class MetricsExporter:
             self.g_devices.labels(rname).set(len(state.device_ids()))
             self.g_healthy.labels(rname).set(len(state.healthy_ids()))
             self.g_last_alloc.labels(rname).set(plugin.last_allocate_s)
+        poller = getattr(m, "_amdsmi", None)
+        if poller is not None:
+            for bdf, dh in poller.last_snapshot.items():
+                for link in dh.xgmi_links:
+                    self.g_xgmi_up.labels(bdf, str(link.index)).set(
+                        1.0 if link.status == "up" else 0.0)
+        if m.topology is not None:
+            for bdf in m.topology.degraded:
+                self.g_hive_degraded.labels(bdf).set(1.0)
 
     def counter_families(self):
         """Counter-typed *_total families, one sample per resource."""
@@ -70,7 +90,16 @@ class MetricsExporter:
             c_allocs.add_metric([rname], plugin.allocations)
             c_fail.add_metric([rname], plugin.allocate_failures)
             c_secs.add_metric([rname], plugin.allocate_seconds_total)
-        return [c_allocs, c_fail, c_secs]
+        c_xgmi_err = CounterMetricFamily(
+            "kxdp_xgmi_link_errors",
+            "Cumulative xGMI link errors per GPU link (amd-smi)",
+            labels=["bdf", "link"])
+        poller = getattr(self.manager, "_amdsmi", None)
+        if poller is not None:
+            for bdf, dh in poller.last_snapshot.items():
+                for link in dh.xgmi_links:
+                    c_xgmi_err.add_metric([bdf, str(link.index)], link.errors)
+        return [c_allocs, c_fail, c_secs, c_xgmi_err]
 
     def start(self, port: int) -> None:
         self._server, _ = start_http_server(port, registry=self.registry)

@@ -125,7 +125,8 @@ class PluginManager:
         if self.cfg.amdsmi_health:
             from ..health.amdsmi_health import AmdSmiPoller
             self._amdsmi = AmdSmiPoller(
-                self.cfg.health_poll_interval_s, self._on_amdsmi_health
+                self.cfg.health_poll_interval_s, self._on_amdsmi_health,
+                on_xgmi=self._on_xgmi_telemetry,
             )
             self._amdsmi.start()
         if self.cfg.gpu_probe_interval_s > 0:
@@ -197,6 +198,22 @@ class PluginManager:
                         log.warning("amd-smi: %s unhealthy: %s", bdf, reasons)
                     state.set_health(gid, healthy, source="amdsmi")
                     return
+
+    def _on_xgmi_telemetry(self, snap) -> None:
+        """Per-link xGMI health → hive degradation:
+        a GPU with link errors or a down link keeps serving but loses its
+        hive membership in placement scoring, so multi-GPU pods prefer
+        GPUs whose fabric is intact."""
+        if self.topology is None:
+            return
+        sick = {bdf for bdf, dh in snap.items() if dh.xgmi_sick}
+        if sick != set(self.topology.degraded):
+            if sick:
+                log.warning("xGMI degraded link(s) on %s — excluded from "
+                            "hive placement", sorted(sick))
+            else:
+                log.info("xGMI links recovered; hive placement restored")
+            self.topology.set_degraded(sick)
 
     def _rescan_loop(self) -> None:
         while not self._stop.wait(self.cfg.rescan_interval_s):
@@ -294,7 +311,9 @@ class PluginManager:
                  len(self.inventory.devices) if self.inventory else 0,
                  len(inv.devices))
         self.inventory = inv
+        prev_degraded = self.topology.degraded if self.topology else frozenset()
         self.topology = load_topology(self.cfg, inv)
+        self.topology.degraded = prev_degraded  # carry link-health across rescan
         self.cdi_spec_path = self._write_cdi(inv)
         grouped = self._group_by_resource(inv)
 

@@ -65,9 +65,21 @@ class GPUTopology:
     numa_of: Dict[str, int] = field(default_factory=dict)   # bdf → numa node
     xgmi_gbps: float = 0.0
     source: str = "none"   # kfd | hint | none
+    # BDFs whose xGMI links are sick (amd-smi link errors / down links):
+    # they score as fabric-less so placement prefers GPUs with healthy
+    # links — without marking the GPU itself Unhealthy (it still computes).
+    degraded: frozenset = frozenset()
 
     def hive(self, bdf: str) -> str:
+        if bdf in self.degraded:
+            return ""
         return self.hive_of.get(bdf, "")
+
+    def set_degraded(self, bdfs) -> None:
+        """Atomically replace the degraded-link set (called from the
+        amd-smi poller thread; readers see old or new set, never a
+        partial one)."""
+        self.degraded = frozenset(b.lower() for b in bdfs)
 
 
 def topology_from_kfd(sysfs_root: str) -> Optional[GPUTopology]:

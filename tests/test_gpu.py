@@ -144,6 +144,26 @@ def test_amdsmi_live_snapshot():
         "temperature missing — amdsmi temp API name/enum likely wrong")
 
 
+def test_amdsmi_live_xgmi_telemetry():
+    """xGMI link telemetry on live silicon: the API-probing ladder in
+    _read_xgmi_links must produce a well-formed (possibly empty on a
+    single-GPU box with no fabric peers) link list, never crash, and any
+    reported link must carry a sane index/status/error triple."""
+    from kata_xpu_device_plugin_amd.health.amdsmi_health import snapshot
+
+    snap = snapshot()
+    assert snap, "amd-smi saw no devices on a GPU box"
+    for dh in snap.values():
+        assert isinstance(dh.xgmi_links, list)
+        for link in dh.xgmi_links:
+            assert link.index >= 0
+            assert link.status in ("up", "down", "disabled", "unknown")
+            assert link.errors >= 0
+        # a healthy lone GPU must not be reported fabric-sick
+        if not dh.xgmi_links:
+            assert not dh.xgmi_sick
+
+
 def test_doctor_live_node():
     """kxdp-doctor on the live box: amdgpu-bound GPUs → pre-provisioning
     verdict (exit 0) with real discovery/topology/amd-smi content."""

@@ -488,3 +488,90 @@ def test_state_watcher_queue_overflow_is_safe(tmp_path):
     snap = dict((d.id, h) for d, h in st.snapshot())
     assert snap["70"] in (True, False)
     st.unwatch(q)
+
+
+# --- xGMI link health → hive degradation ---------------
+
+def test_xgmi_degradation_steers_placement(tmp_path):
+    """A GPU with sick xGMI links loses hive membership in placement: a
+    2-GPU pod avoids it while intact-fabric peers exist — without the GPU
+    going Unhealthy."""
+    from kata_xpu_device_plugin_amd.health.amdsmi_health import (
+        DeviceHealth, XgmiLinkHealth)
+    from kata_xpu_device_plugin_amd.topology.hive import preferred_allocation
+
+    # two 2-GPU hives
+    node = make_mock_node(str(tmp_path), n_gpus=4, hives=[[0, 1], [2, 3]],
+                          kfd=False, hint=True)
+    mgr = PluginManager(node.config())
+    mgr.setup()
+    inv, topo = mgr.inventory, mgr.topology
+    bdfs = sorted(d.primary.bdf for d in inv.devices.values())
+
+    def pick2():
+        return sorted(preferred_allocation(
+            topo, inv, sorted(inv.devices), [], 2))
+
+    healthy_pick = pick2()
+    assert len(healthy_pick) == 2
+
+    # GPU 0 (hive 1) reports link errors → degrade
+    sick_bdf = bdfs[0]
+    snap = {sick_bdf: DeviceHealth(
+        bdf=sick_bdf,
+        xgmi_links=[XgmiLinkHealth(index=0, status="up", errors=3)])}
+    mgr._on_xgmi_telemetry(snap)
+    assert topo.degraded == frozenset({sick_bdf})
+
+    pick = pick2()
+    chosen_bdfs = {inv.devices[g].primary.bdf for g in pick}
+    assert sick_bdf not in chosen_bdfs, \
+        "placement must avoid the degraded-fabric GPU"
+    # health is NOT affected — the GPU still allocates
+    for st_ in mgr.states.values():
+        assert all(h for _, h in st_.snapshot())
+
+    # recovery restores the hive
+    mgr._on_xgmi_telemetry({sick_bdf: DeviceHealth(bdf=sick_bdf)})
+    assert topo.degraded == frozenset()
+
+
+def test_xgmi_metrics_exported(tmp_path):
+    from prometheus_client import generate_latest
+    from kata_xpu_device_plugin_amd.metrics import MetricsExporter
+    from kata_xpu_device_plugin_amd.health.amdsmi_health import (
+        AmdSmiPoller, DeviceHealth, XgmiLinkHealth)
+
+    node = make_mock_node(str(tmp_path), n_gpus=1, kfd=False, hint=False)
+    mgr = PluginManager(node.config())
+    mgr.setup()
+    snap = {"0000:0a:00.0": DeviceHealth(
+        bdf="0000:0a:00.0",
+        xgmi_links=[XgmiLinkHealth(index=0, status="up", errors=0),
+                    XgmiLinkHealth(index=1, status="down", errors=2)])}
+    mgr._amdsmi = AmdSmiPoller(999, lambda *a: None,
+                               snapshot_fn=lambda: snap,
+                               on_xgmi=mgr._on_xgmi_telemetry)
+    mgr._amdsmi.poll_once()
+    exp = MetricsExporter(mgr)
+    text = generate_latest(exp.registry).decode()
+    assert 'kxdp_xgmi_link_up{bdf="0000:0a:00.0",link="0"} 1.0' in text
+    assert 'kxdp_xgmi_link_up{bdf="0000:0a:00.0",link="1"} 0.0' in text
+    assert "# TYPE kxdp_xgmi_link_errors_total counter" in text
+    assert 'kxdp_xgmi_link_errors_total{bdf="0000:0a:00.0",link="1"} 2.0' in text
+    assert 'kxdp_xgmi_degraded{bdf="0000:0a:00.0"} 1.0' in text
+
+
+def test_xgmi_degradation_survives_rescan(tmp_path):
+    from kata_xpu_device_plugin_amd.health.amdsmi_health import (
+        DeviceHealth, XgmiLinkHealth)
+    node = make_mock_node(str(tmp_path), n_gpus=2, kfd=False, hint=True)
+    mgr = PluginManager(node.config())
+    mgr.setup()
+    bdf = sorted(d.primary.bdf for d in mgr.inventory.devices.values())[0]
+    mgr._on_xgmi_telemetry({bdf: DeviceHealth(
+        bdf=bdf, xgmi_links=[XgmiLinkHealth(index=0, errors=1)])})
+    assert bdf in mgr.topology.degraded
+    node.add_gpu(MockGPU(bdf="0000:66:00.0", iommu_group="99"))
+    assert mgr.rescan() is True
+    assert bdf in mgr.topology.degraded, "degradation lost across rescan"
