@@ -24,6 +24,12 @@ STRATEGIES = (STRATEGY_CDI_CRI, STRATEGY_CDI_ANNOTATIONS, STRATEGY_DEVICE_NODES)
 DEFAULT_VENDOR_ALLOWLIST = (0x1002,)  # AMD/ATI GPUs (AMD CPU-side IP is 0x1022)
 
 
+def bundled_pci_ids_path() -> str:
+    """Pinned pci.ids snapshot shipped inside the package."""
+    return os.path.join(os.path.dirname(os.path.abspath(__file__)),
+                        "data", "pci.ids")
+
+
 def _env(name: str, default: str) -> str:
     return os.environ.get(f"KXDP_{name}", default)
 
@@ -75,11 +81,16 @@ class Config:
 
     # --- naming ---
     resource_namespace: str = field(default_factory=lambda: _env("NAMESPACE", "amd.com"))
+    # pci.ids fallback chain, tried in order after the curated Instinct
+    # table. First entry: the BUNDLED pinned snapshot (data/pci.ids) so an
+    # air-gapped node without a distro pci.ids still names known silicon
+    # deterministically; host copies follow for anything missing there.
     pci_ids_paths: Sequence[str] = field(
         default_factory=lambda: tuple(
             p for p in _env(
                 "PCI_IDS",
-                "/usr/share/misc/pci.ids:/usr/share/hwdata/pci.ids:/usr/pci.ids",
+                bundled_pci_ids_path()
+                + ":/usr/share/misc/pci.ids:/usr/share/hwdata/pci.ids:/usr/pci.ids",
             ).split(":")
             if p
         )

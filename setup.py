@@ -55,6 +55,10 @@ if __name__ == "__main__" or "setuptools" in sys.modules:
                 f"{PKG}.plugin", f"{PKG}.health", f"{PKG}.utils",
                 f"{PKG}.testing", f"{PKG}.tools",
             ],
+            # pinned pci.ids snapshot: deterministic naming on air-gapped
+            # nodes (first entry of config.pci_ids_paths)
+            package_data={PKG: ["data/pci.ids"]},
+            include_package_data=True,
             entry_points={
                 "console_scripts": [
                     "kata-xpu-device-plugin-amd=kata_xpu_device_plugin_amd.plugin.manager:main",
