@@ -27,6 +27,11 @@ log = get_logger(__name__)
 # GPU is unhealthy if any of these RAS blocks report uncorrectable errors.
 _FATAL_TEMP_C = 105.0
 
+# RAS IP blocks worth attributing faults to (AmdSmiGpuBlock names):
+# UMC = HBM memory controller, GFX = compute, SDMA = copy engines,
+# XGMI_WAFL = fabric, MMHUB = memory hub, PCIE_BIF = host link.
+_RAS_BLOCKS = ("UMC", "GFX", "SDMA", "XGMI_WAFL", "MMHUB", "PCIE_BIF")
+
 
 @dataclass
 class XgmiLinkHealth:
@@ -47,6 +52,9 @@ class DeviceHealth:
     # placement (topology/hive.py::GPUTopology.degraded) and is exported
     # per-link to Prometheus.
     xgmi_links: List[XgmiLinkHealth] = field(default_factory=list)
+    # Per-RAS-block uncorrectable counts (fault ATTRIBUTION — which IP
+    # block is dying: HBM controller vs compute vs fabric), best-effort.
+    ecc_by_block: Dict[str, int] = field(default_factory=dict)
 
     @property
     def xgmi_sick(self) -> bool:
@@ -149,6 +157,27 @@ def snapshot() -> Dict[str, DeviceHealth]:
                 if ue > 0:
                     dh.healthy = False
                     dh.reasons.append(f"{ue} uncorrectable ECC errors")
+            except Exception:
+                pass
+            # Per-block attribution (which IP block is failing); also a
+            # fallback verdict when the totals API is unavailable.
+            try:
+                blocks = getattr(amdsmi, "AmdSmiGpuBlock", None)
+                for bname in _RAS_BLOCKS if blocks is not None else ():
+                    blk = getattr(blocks, bname, None)
+                    if blk is None:
+                        continue
+                    try:
+                        c = amdsmi.amdsmi_get_gpu_ecc_count(h, blk)
+                        bue = int(c.get("uncorrectable_count", 0) or 0)
+                    except Exception:
+                        continue
+                    dh.ecc_by_block[bname] = bue
+                    if bue > 0:
+                        dh.healthy = False
+                        reason = f"{bue} uncorrectable in {bname}"
+                        if reason not in dh.reasons:
+                            dh.reasons.append(reason)
             except Exception:
                 pass
             # Thermals (edge/junction)

@@ -94,12 +94,18 @@ class MetricsExporter:
             "kxdp_xgmi_link_errors",
             "Cumulative xGMI link errors per GPU link (amd-smi)",
             labels=["bdf", "link"])
+        c_ecc = CounterMetricFamily(
+            "kxdp_ecc_uncorrectable",
+            "Uncorrectable RAS errors per IP block (amd-smi)",
+            labels=["bdf", "block"])
         poller = getattr(self.manager, "_amdsmi", None)
         if poller is not None:
             for bdf, dh in poller.last_snapshot.items():
                 for link in dh.xgmi_links:
                     c_xgmi_err.add_metric([bdf, str(link.index)], link.errors)
-        return [c_allocs, c_fail, c_secs, c_xgmi_err]
+                for block, count in dh.ecc_by_block.items():
+                    c_ecc.add_metric([bdf, block], count)
+        return [c_allocs, c_fail, c_secs, c_xgmi_err, c_ecc]
 
     def start(self, port: int) -> None:
         self._server, _ = start_http_server(port, registry=self.registry)

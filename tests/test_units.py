@@ -575,3 +575,24 @@ def test_xgmi_degradation_survives_rescan(tmp_path):
     node.add_gpu(MockGPU(bdf="0000:66:00.0", iommu_group="99"))
     assert mgr.rescan() is True
     assert bdf in mgr.topology.degraded, "degradation lost across rescan"
+
+
+def test_ecc_block_attribution_metrics(tmp_path):
+    from prometheus_client import generate_latest
+    from kata_xpu_device_plugin_amd.metrics import MetricsExporter
+    from kata_xpu_device_plugin_amd.health.amdsmi_health import (
+        AmdSmiPoller, DeviceHealth)
+
+    node = make_mock_node(str(tmp_path), n_gpus=1, kfd=False, hint=False)
+    mgr = PluginManager(node.config())
+    mgr.setup()
+    snap = {"0000:0a:00.0": DeviceHealth(
+        bdf="0000:0a:00.0", healthy=False,
+        reasons=["2 uncorrectable in UMC"],
+        ecc_by_block={"UMC": 2, "GFX": 0})}
+    mgr._amdsmi = AmdSmiPoller(999, lambda *a: None, snapshot_fn=lambda: snap)
+    mgr._amdsmi.poll_once()
+    text = generate_latest(MetricsExporter(mgr).registry).decode()
+    assert "# TYPE kxdp_ecc_uncorrectable_total counter" in text
+    assert 'kxdp_ecc_uncorrectable_total{bdf="0000:0a:00.0",block="UMC"} 2.0' in text
+    assert 'kxdp_ecc_uncorrectable_total{bdf="0000:0a:00.0",block="GFX"} 0.0' in text
