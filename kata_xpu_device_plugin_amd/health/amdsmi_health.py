@@ -72,17 +72,26 @@ def _read_xgmi_links(amdsmi, h) -> List[XgmiLinkHealth]:
     fall back to gpu_metrics' xgmi_link_status array; every call is
     defensive (missing API ⇒ fewer fields, never an exception out)."""
     links: List[XgmiLinkHealth] = []
-    # 1) per-link status (amdsmi_get_gpu_xgmi_link_status, ROCm >= 6.3)
+
+    def status_name(v) -> str:
+        name = str(v).rsplit(".", 1)[-1].strip().lower()
+        if name.isdigit():  # numeric enum: 0=disabled,1=up,2=down (smi)
+            name = {0: "disabled", 1: "up", 2: "down"}.get(int(name),
+                                                           "unknown")
+        return name if name in ("up", "down", "disabled") else "unknown"
+
+    # 1) per-link status (amdsmi_get_gpu_xgmi_link_status, ROCm >= 6.3).
+    # Return shape varies by release: list of enums, dict {"status": [..]}
+    # or a single scalar/string — only a real sequence is per-link
+    # (live-MI355X lesson: a str here would iterate as characters).
     try:
         st = amdsmi.amdsmi_get_gpu_xgmi_link_status(h)
-        # dict {"status": [..]} or plain list depending on version
         raw = st.get("status") if isinstance(st, dict) else st
-        for i, v in enumerate(raw or []):
-            name = str(v).rsplit(".", 1)[-1].lower()
-            if name.isdigit():  # numeric enum: 0=disabled,1=up,2=down (smi)
-                name = {0: "disabled", 1: "up", 2: "down"}.get(int(name),
-                                                               "unknown")
-            links.append(XgmiLinkHealth(index=i, status=name))
+        if isinstance(raw, (list, tuple)):
+            for i, v in enumerate(raw):
+                links.append(XgmiLinkHealth(index=i, status=status_name(v)))
+        elif raw is not None and not isinstance(raw, (str, bytes)):
+            links.append(XgmiLinkHealth(index=0, status=status_name(raw)))
     except Exception:
         pass
     if not links:
@@ -90,6 +99,8 @@ def _read_xgmi_links(amdsmi, h) -> List[XgmiLinkHealth]:
         try:
             gm = amdsmi.amdsmi_get_gpu_metrics_info(h)
             raw = gm.get("xgmi_link_status", []) if isinstance(gm, dict) else []
+            if not isinstance(raw, (list, tuple)):
+                raw = []
             for i, v in enumerate(raw):
                 try:
                     iv = int(v)
