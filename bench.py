@@ -131,7 +131,8 @@ def main() -> int:
     # --- every rank: its own channel to the plugin ----------------------
     import grpc
     channel = grpc.insecure_channel(
-        f"unix://{os.path.join(endpoint_dir, endpoint)}")
+        f"unix://{os.path.join(endpoint_dir, endpoint)}",
+        options=[("grpc.optimization_target", "latency")])
     grpc.channel_ready_future(channel).result(timeout=10)
     plugin = api.DevicePluginStub(channel)
 

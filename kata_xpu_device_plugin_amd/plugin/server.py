@@ -367,7 +367,10 @@ class XPUDevicePlugin:
 
         async def _main():
             server = grpc.aio.server(
-                options=[("grpc.max_concurrent_streams", 64)])
+                options=[("grpc.max_concurrent_streams", 64),
+                         # admission RPCs are µs-scale: bias grpc-core
+                         # buffering/batching toward latency
+                         ("grpc.optimization_target", "latency")])
             api.add_device_plugin_servicer(server, self)
             # The socket dir normally exists (kubelet / hostPath
             # DirectoryOrCreate); create it ourselves so a daemon started

@@ -107,12 +107,21 @@ def diagnose(cfg: Config, probe: bool = False) -> dict:
     smi = amdsmi_snapshot()
     doc["amdsmi"] = {
         b: {"healthy": h.healthy, "temp_c": h.temperature_c,
-            "uncorrectable_ecc": h.uncorrectable_errors, "reasons": h.reasons}
+            "uncorrectable_ecc": h.uncorrectable_errors, "reasons": h.reasons,
+            "ecc_by_block": dict(h.ecc_by_block),
+            "xgmi_links": [
+                {"index": l.index, "status": l.status, "errors": l.errors}
+                for l in h.xgmi_links
+            ]}
         for b, h in smi.items()
     }
     for b, h in smi.items():
         if not h.healthy:
             doc["problems"].append(f"amd-smi: {b} unhealthy: {h.reasons}")
+        if h.xgmi_sick:
+            doc["problems"].append(
+                f"amd-smi: {b} xGMI link degraded "
+                f"(placement will avoid this GPU's hive membership)")
 
     if probe:
         try:
