@@ -233,3 +233,27 @@ def test_cdi_read_spec_never_crashes_outside_valueerror(raw):
         import yaml as _yaml
         assert isinstance(e, (_yaml.YAMLError, ValueError, KeyError,
                               TypeError, AttributeError)), type(e)
+
+
+# --- CDI schema validator fuzz ----------------------------------------------
+
+_json_values = st.recursive(
+    st.one_of(st.none(), st.booleans(), st.integers(), st.text(max_size=8)),
+    lambda children: st.one_of(
+        st.lists(children, max_size=4),
+        st.dictionaries(st.text(max_size=8), children, max_size=4)),
+    max_leaves=12,
+)
+
+
+@given(_json_values)
+@settings(max_examples=200, deadline=None)
+def test_cdi_schema_validator_never_raises(obj):
+    """validate_spec_obj on arbitrary JSON-shaped input returns an error
+    list — it must never throw (it guards the write path)."""
+    from kata_xpu_device_plugin_amd.cdi.schema import validate_spec_obj
+    errors = validate_spec_obj(obj)
+    assert isinstance(errors, list)
+    if errors == []:
+        # only a structurally valid CDI spec validates cleanly
+        assert isinstance(obj, dict) and "cdiVersion" in obj
