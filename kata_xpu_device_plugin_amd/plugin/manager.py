@@ -252,8 +252,24 @@ class PluginManager:
             log.info("signal %d — shutting down", signum)
             self._stop.set()
 
+        def _hup(signum, frame):
+            # operator-triggered live rescan (e.g. right after enabling
+            # SR-IOV VFs) — run off the signal frame; rescan() serializes
+            # itself via _rescan_lock
+            log.info("SIGHUP — rescanning inventory")
+            threading.Thread(target=self._safe_rescan, name="kxdp-hup-rescan",
+                             daemon=True).start()
+
         signal.signal(signal.SIGTERM, _sig)
         signal.signal(signal.SIGINT, _sig)
+        signal.signal(signal.SIGHUP, _hup)
+
+    def _safe_rescan(self) -> None:
+        try:
+            if not self._stop.is_set():
+                self.rescan()
+        except Exception:
+            log.exception("SIGHUP rescan failed")
 
     def run_forever(self) -> None:
         """Block until SIGTERM/SIGINT (the reference blocks on a channel it

@@ -441,6 +441,51 @@ def test_daemon_survives_missing_kubelet(tmp_path):
             stub.stop()
 
 
+def test_daemon_main_sighup_rescan(tmp_path):
+    """SIGHUP triggers a live rescan in the real daemon process: a GPU
+    added after startup becomes allocatable without restart."""
+    import signal
+    import subprocess
+    import sys
+
+    node = make_mock_node(str(tmp_path), n_gpus=2)
+    cfg = node.config()
+    stub = KubeletStub(cfg.kubelet_socket_dir)
+    stub.start()
+    env = dict(os.environ,
+               KXDP_SYSFS_ROOT=cfg.sysfs_root,
+               KXDP_DEV_ROOT=cfg.dev_root,
+               KXDP_CDI_DIR=cfg.cdi_dir,
+               KXDP_KUBELET_DIR=cfg.kubelet_socket_dir,
+               KXDP_TOPOLOGY_HINT=cfg.topology_hint_path,
+               KXDP_AMDSMI_HEALTH="false",
+               PYTHONPATH=os.path.dirname(os.path.dirname(
+                   os.path.abspath(__file__))))
+    proc = subprocess.Popen(
+        [sys.executable, "-m", "kata_xpu_device_plugin_amd"],
+        env=env, stdout=subprocess.PIPE, stderr=subprocess.PIPE, text=True)
+    try:
+        regs = stub.wait_for_registration(1, timeout=20)
+        ps = stub.plugin_stub(regs[0].endpoint)
+        stream = ps.ListAndWatch(api.Empty())
+        assert len(next(stream).devices) == 2
+
+        node.add_gpu(MockGPU(bdf="0000:77:00.0", iommu_group="99"))
+        proc.send_signal(signal.SIGHUP)
+        upd = next(stream)   # rescan pushes the new inventory
+        assert sorted(d.id for d in upd.devices) == ["70", "71", "99"]
+        stream.cancel()
+        resp = ps.Allocate(api.AllocateRequest(container_requests=[
+            api.ContainerAllocateRequest(devices_ids=["99"])]))
+        assert resp.container_responses[0].cdi_devices[0].name == "amd.com/gpu=99"
+        proc.send_signal(signal.SIGTERM)
+        assert proc.wait(timeout=15) == 0
+    finally:
+        if proc.poll() is None:
+            proc.kill()
+        stub.stop()
+
+
 def test_daemon_main_sigterm(tmp_path):
     """End-to-end daemon process: `python -m kata_xpu_device_plugin_amd`
     starts against a mock node, serves, and shuts down cleanly on SIGTERM
