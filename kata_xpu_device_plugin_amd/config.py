@@ -145,6 +145,12 @@ class Config:
     # 0 = off. Rescans that find no change are no-ops (no CDI rewrite).
     rescan_interval_s: float = field(default_factory=lambda: _env_float("RESCAN_S", 0.0))
 
+    # --- diagnostics ---
+    # When set, every plugin server records per-RPC handler wall time and
+    # event-loop scheduling lag, and dumps a JSON breakdown to this path
+    # on shutdown (tail-latency attribution; zero cost when unset).
+    rpc_timing_path: str = field(default_factory=lambda: _env("RPC_TIMING", ""))
+
     # --- performance ---
     # Freeze the startup object graph + raise GC thresholds after start()
     # (kills multi-ms generational-GC pauses in Allocate's p99 tail).

@@ -102,6 +102,9 @@ class XPUDevicePlugin:
         self._vendor_list = list(cfg.vendor_allowlist)
         self._qn_cache: Dict[str, str] = {}      # gid → "kind=gid"
         self._bdf_cache = None                   # (generation, gid→bdf)
+        # optional tail-latency attribution (config.rpc_timing_path)
+        self._timing = {"allocate_us": [], "preferred_us": [],
+                        "loop_lag_us": []} if cfg.rpc_timing_path else None
         self._env_res_name = ENV_PCI_RESOURCE_PREFIX + resource_name.upper(
         ).replace("/", "_").replace(".", "_").replace("-", "_")
 
@@ -262,9 +265,12 @@ class XPUDevicePlugin:
         self.allocations += 1
         self.last_allocate_s = time.perf_counter() - t0
         self.allocate_seconds_total += self.last_allocate_s
+        if self._timing is not None:
+            self._timing["allocate_us"].append(self.last_allocate_s * 1e6)
         return response
 
     async def GetPreferredAllocation(self, request, context):
+        t0 = time.perf_counter() if self._timing is not None else 0.0
         resp = api.PreferredAllocationResponse()
         # gid→BDF map cached per state generation: rebuilt only when the
         # device set or health actually changed, not per admission RPC
@@ -283,6 +289,9 @@ class XPUDevicePlugin:
                 self.topo, bdf_of, available, must, creq.allocation_size
             )
             resp.container_responses.add(device_ids=pick)
+        if self._timing is not None:
+            self._timing["preferred_us"].append(
+                (time.perf_counter() - t0) * 1e6)
         return resp
 
     async def PreStartContainer(self, request, context):
@@ -398,13 +407,30 @@ class XPUDevicePlugin:
             await server.start()
             self._aio_server = server
             self._stop_async = asyncio.Event()
+            lag_task = None
+            if self._timing is not None:
+                async def _lag_monitor():
+                    # scheduling lag: how late a 1 ms sleep actually wakes
+                    # — queueing other work inflicts on the loop.
+                    while not self._stop_async.is_set():
+                        t0 = time.perf_counter()
+                        await asyncio.sleep(0.001)
+                        lag = (time.perf_counter() - t0 - 0.001) * 1e6
+                        self._timing["loop_lag_us"].append(max(0.0, lag))
+                        if len(self._timing["loop_lag_us"]) > 200_000:
+                            del self._timing["loop_lag_us"][:100_000]
+                lag_task = asyncio.ensure_future(_lag_monitor())
             self._loop_ready.set()
             # The loop thread owns the full server lifetime: it waits for
             # the cross-thread stop signal and performs the (async) stop
             # itself — run_until_complete would otherwise tear the loop
             # down with the stop coroutine still pending.
             await self._stop_async.wait()
+            if lag_task is not None:
+                lag_task.cancel()
             await server.stop(grace=1.0)
+            if self._timing is not None:
+                self._dump_timing()
 
         try:
             loop.run_until_complete(_main())
@@ -416,6 +442,31 @@ class XPUDevicePlugin:
                 loop.close()
             except Exception:
                 pass
+
+    def _dump_timing(self) -> None:
+        import json
+        import statistics as stats
+
+        def dist(xs):
+            if not xs:
+                return None
+            xs = sorted(xs)
+            n = len(xs)
+            return {"n": n,
+                    "p50": round(xs[n // 2], 1),
+                    "p90": round(xs[min(n - 1, int(n * 0.9))], 1),
+                    "p99": round(xs[min(n - 1, int(n * 0.99))], 1),
+                    "max": round(xs[-1], 1),
+                    "mean": round(stats.fmean(xs), 1)}
+
+        out = {k: dist(v) for k, v in self._timing.items()}
+        path = f"{self.cfg.rpc_timing_path}.{self.socket_name}.json"
+        try:
+            with open(path, "w") as f:
+                json.dump(out, f, indent=1)
+            log.info("rpc timing breakdown → %s", path)
+        except OSError as e:
+            log.warning("could not write rpc timing: %s", e)
 
     def register_with_kubelet(self) -> None:
         """Register against kubelet.sock (reference Register :200-219)."""

@@ -596,3 +596,37 @@ def test_ecc_block_attribution_metrics(tmp_path):
     assert "# TYPE kxdp_ecc_uncorrectable_total counter" in text
     assert 'kxdp_ecc_uncorrectable_total{bdf="0000:0a:00.0",block="UMC"} 2.0' in text
     assert 'kxdp_ecc_uncorrectable_total{bdf="0000:0a:00.0",block="GFX"} 0.0' in text
+
+
+def test_rpc_timing_breakdown(tmp_path):
+    """KXDP_RPC_TIMING: per-RPC handler durations + loop lag dumped as a
+    JSON distribution on shutdown (tail attribution)."""
+    import json, glob
+    node = make_mock_node(str(tmp_path), n_gpus=2)
+    cfg = node.config(rpc_timing_path=os.path.join(str(tmp_path), "timing"))
+    stub = KubeletStub(cfg.kubelet_socket_dir)
+    stub.start()
+    mgr = PluginManager(cfg)
+    mgr.setup()
+    mgr.start()
+    try:
+        reg = stub.wait_for_registration(1)[0]
+        ps = stub.plugin_stub(reg.endpoint)
+        for _ in range(5):
+            pref = ps.GetPreferredAllocation(api.PreferredAllocationRequest(
+                container_requests=[api.ContainerPreferredAllocationRequest(
+                    available_device_ids=["70", "71"], allocation_size=1)]))
+            ps.Allocate(api.AllocateRequest(container_requests=[
+                api.ContainerAllocateRequest(
+                    devices_ids=list(pref.container_responses[0].device_ids))]))
+        time.sleep(0.05)
+    finally:
+        mgr.stop()
+        stub.stop()
+    files = glob.glob(os.path.join(str(tmp_path), "timing.*.json"))
+    assert files, "timing dump missing"
+    doc = json.load(open(files[0]))
+    assert doc["allocate_us"]["n"] == 5
+    assert doc["preferred_us"]["n"] == 5
+    assert doc["allocate_us"]["p50"] > 0
+    assert doc["loop_lag_us"] is None or doc["loop_lag_us"]["n"] >= 1
