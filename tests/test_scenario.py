@@ -91,11 +91,36 @@ def test_full_lifecycle(tmp_path):
         mgr.rescan()
         spec = read_spec(mgr.cdi_spec_path)
         assert {"200", "201"} <= set(spec.device_names())
-        # VFs are a NEW resource name → flagged for restart (logged), but
-        # existing MI355X plugin keeps serving
+        # VFs are a NEW resource name → its plugin starts and registers IN
+        # PLACE (round 2: no daemon restart), and the existing MI355X
+        # plugin keeps serving
+        vf_reg = next(r for r in stub.wait_for_registration(2)
+                      if r.resource_name.endswith("_VF"))
+        ps_vf = stub.plugin_stub(vf_reg.endpoint)
+        resp = ps_vf.Allocate(api.AllocateRequest(container_requests=[
+            api.ContainerAllocateRequest(devices_ids=["200"])]))
+        assert resp.container_responses[0].cdi_devices[0].name == "amd.com/gpu=200"
         resp = ps.Allocate(api.AllocateRequest(container_requests=[
             api.ContainerAllocateRequest(devices_ids=["70"])]))
         assert resp.container_responses[0].cdi_devices[0].name == "amd.com/gpu=70"
+
+        # ---- xGMI link sickness: placement avoids the GPU, health intact --
+        from kata_xpu_device_plugin_amd.health.amdsmi_health import (
+            DeviceHealth, XgmiLinkHealth)
+        sick_bdf = mgr.inventory.devices["71"].primary.bdf   # hive 1
+        mgr._on_xgmi_telemetry({sick_bdf: DeviceHealth(
+            bdf=sick_bdf,
+            xgmi_links=[XgmiLinkHealth(index=3, status="down", errors=1)])})
+        pref = ps.GetPreferredAllocation(api.PreferredAllocationRequest(
+            container_requests=[api.ContainerPreferredAllocationRequest(
+                available_device_ids=ids, allocation_size=2)]))
+        pick = list(pref.container_responses[0].device_ids)
+        assert "71" not in pick, "degraded-fabric GPU must be avoided"
+        # ... but it still allocates when kubelet insists (health intact)
+        resp = ps.Allocate(api.AllocateRequest(container_requests=[
+            api.ContainerAllocateRequest(devices_ids=["71"])]))
+        assert resp.container_responses[0].cdi_devices[0].name == "amd.com/gpu=71"
+        mgr._on_xgmi_telemetry({sick_bdf: DeviceHealth(bdf=sick_bdf)})
 
         # ---- device failure: vfio node vanishes → Unhealthy + reject ----
         stream = ps.ListAndWatch(api.Empty())
