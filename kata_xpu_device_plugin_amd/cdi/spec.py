@@ -107,11 +107,12 @@ class CDISpec:
                 raise ValueError(f"CDI device {d.name!r} has no device nodes")
 
 
-def build_spec(inv: NodeInventory, kind: str, dev_root: str = "/dev") -> CDISpec:
+def build_spec(inv: NodeInventory, kind: str, dev_root: str = "/dev",
+               cdi_version: str = CDI_VERSION) -> CDISpec:
     """NodeInventory → CDISpec. Reference analog: generateCDISpec
     (`device_plugin.go:55-80`), with per-group devices instead of
     per-function."""
-    spec = CDISpec(kind=kind)
+    spec = CDISpec(kind=kind, cdi_version=cdi_version)
     for gid in inv.device_ids():
         dev = inv.devices[gid]
         spec.devices.append(device_entry(dev, kind, dev_root))

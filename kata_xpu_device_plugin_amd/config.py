@@ -105,6 +105,10 @@ class Config:
     cdi_kind: str = field(default_factory=lambda: _env("CDI_KIND", "amd.com/gpu"))
     cdi_spec_name: str = field(default_factory=lambda: _env("CDI_SPEC_NAME", "kxdp-vfio"))
     cdi_format: str = field(default_factory=lambda: _env("CDI_FORMAT", "yaml"))  # yaml|json
+    # Emitted cdiVersion. 0.8.0 default; turn down for runtimes predating
+    # it (the reference wrote frozen 0.6.0, cdi/spec.go:12 — everything
+    # this plugin emits is already expressible at 0.6.0).
+    cdi_version: str = field(default_factory=lambda: _env("CDI_VERSION", "0.8.0"))
     device_list_strategy: str = field(default_factory=lambda: _env("STRATEGY", STRATEGY_CDI_CRI))
 
     # --- kubelet ---
@@ -173,6 +177,11 @@ class Config:
             )
         if self.cdi_format not in ("yaml", "json"):
             raise ValueError(f"cdi_format must be yaml|json, got {self.cdi_format!r}")
+        from .cdi.schema import KNOWN_CDI_VERSIONS
+        if self.cdi_version not in KNOWN_CDI_VERSIONS:
+            raise ValueError(
+                f"cdi_version must be one of {KNOWN_CDI_VERSIONS}, "
+                f"got {self.cdi_version!r}")
         if "/" not in self.cdi_kind:
             raise ValueError(f"cdi_kind must look like vendor/class, got {self.cdi_kind!r}")
         if self.native not in ("auto", "require", "off"):

@@ -73,7 +73,8 @@ class PluginManager:
                         pass
             log.info("no devices: no CDI spec written")
             return None
-        spec = build_spec(inv, self.cfg.cdi_kind, self.cfg.dev_root)
+        spec = build_spec(inv, self.cfg.cdi_kind, self.cfg.dev_root,
+                          self.cfg.cdi_version)
         return write_spec(
             spec, self.cfg.cdi_dir, self.cfg.cdi_spec_name, self.cfg.cdi_format
         )
@@ -340,8 +341,15 @@ class PluginManager:
                 self.plugins[rname].topo = self.topology
 
         # Retire plugins whose resource has no devices left (kubelet drops
-        # the resource when the plugin's socket goes away).
-        for rname in [r for r in self.plugins if r not in grouped]:
+        # the resource when the plugin's socket goes away). Drop their
+        # socket names from the watcher FIRST so the self-inflicted socket
+        # removal isn't mistaken for an external wipe.
+        retiring = [r for r in self.plugins if r not in grouped]
+        if retiring and self.watcher is not None:
+            self.watcher.plugin_socket_names = {
+                p.socket_name for r, p in self.plugins.items()
+                if r not in retiring}
+        for rname in retiring:
             plugin = self.plugins.pop(rname)
             self.states.pop(rname, None)
             try:

@@ -69,7 +69,7 @@ def diagnose(cfg: Config, probe: bool = False) -> dict:
 
     if inv.devices:
         cdi_dir = tempfile.mkdtemp(prefix="kxdp-doctor-")
-        spec = build_spec(inv, cfg.cdi_kind, cfg.dev_root)
+        spec = build_spec(inv, cfg.cdi_kind, cfg.dev_root, cfg.cdi_version)
         write_spec(spec, cdi_dir, cfg.cdi_spec_name, cfg.cdi_format)
         resolver = CDIResolver(cdi_dir)
         cdi_report = []

@@ -35,7 +35,7 @@ def main(argv=None) -> int:
         return 1
     cdi_dir = cfg.cdi_dir if os.access(cfg.cdi_dir, os.W_OK) else \
         tempfile.mkdtemp(prefix="kxdp-validate-")
-    spec = build_spec(inv, cfg.cdi_kind, cfg.dev_root)
+    spec = build_spec(inv, cfg.cdi_kind, cfg.dev_root, cfg.cdi_version)
     spec_file = write_spec(spec, cdi_dir, cfg.cdi_spec_name, cfg.cdi_format)
     from ..cdi.schema import validate_spec_file
     schema_problems = validate_spec_file(spec_file)

@@ -224,3 +224,21 @@ def test_kata_runtime_consumption_contract(tmp_path):
     # container edit the runtime applies before VM boot
     nodes = [n["path"] for n in dev["containerEdits"]["deviceNodes"]]
     assert nodes == [os.path.join(cfg.dev_root, "vfio", "70")]
+
+
+def test_cdi_version_knob(tmp_path, monkeypatch):
+    """KXDP_CDI_VERSION lets operators emit 0.6.0 specs for runtimes
+    predating 0.8.0 (the reference's frozen version, cdi/spec.go:12);
+    unknown versions are rejected at config validation."""
+    import pytest as _pytest
+    from kata_xpu_device_plugin_amd.config import Config
+    node = make_mock_node(str(tmp_path), n_gpus=1, kfd=False, hint=False)
+    cfg = node.config(cdi_version="0.6.0")
+    spec = build_spec(scan_node(cfg), "amd.com/gpu", cfg.dev_root,
+                      cfg.cdi_version)
+    path = write_spec(spec, str(tmp_path), "s", "yaml")
+    doc = yaml.safe_load(open(path))
+    assert doc["cdiVersion"] == "0.6.0"
+    _schema_check(path)
+    with _pytest.raises(ValueError):
+        node.config(cdi_version="9.9.9")
