@@ -159,6 +159,10 @@ class Config:
     # Freeze the startup object graph + raise GC thresholds after start()
     # (kills multi-ms generational-GC pauses in Allocate's p99 tail).
     gc_tuning: bool = field(default_factory=lambda: _env_bool("GC_TUNING", True))
+    # Pin the daemon to a CPU set, e.g. "2-3" or "0,4" (empty = no pin).
+    # On busy nodes this keeps admission latency off noisy-neighbor cores;
+    # matches a DaemonSet cpuset without needing static CPU policy.
+    cpu_affinity: str = field(default_factory=lambda: _env("CPU_AFFINITY", ""))
 
     # --- observability ---
     metrics_port: int = field(default_factory=lambda: _env_int("METRICS_PORT", 0))  # 0 = off

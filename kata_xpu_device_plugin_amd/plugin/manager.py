@@ -146,6 +146,28 @@ class PluginManager:
             self._metrics.start(self.cfg.metrics_port)
         if self.cfg.gc_tuning:
             self._tune_gc()
+        if self.cfg.cpu_affinity:
+            self._pin_cpus(self.cfg.cpu_affinity)
+
+    @staticmethod
+    def _pin_cpus(spec: str) -> None:
+        """Pin this process (all threads) to the given CPU list spec."""
+        cpus = set()
+        try:
+            for part in spec.split(","):
+                part = part.strip()
+                if not part:
+                    continue
+                if "-" in part:
+                    lo, hi = part.split("-", 1)
+                    cpus.update(range(int(lo), int(hi) + 1))
+                else:
+                    cpus.add(int(part))
+            if cpus:
+                os.sched_setaffinity(0, cpus)
+                log.info("pinned daemon to CPUs %s", sorted(cpus))
+        except (ValueError, OSError) as e:
+            log.warning("cpu_affinity %r not applied: %s", spec, e)
 
     @staticmethod
     def _tune_gc() -> None:

@@ -630,3 +630,20 @@ def test_rpc_timing_breakdown(tmp_path):
     assert doc["preferred_us"]["n"] == 5
     assert doc["allocate_us"]["p50"] > 0
     assert doc["loop_lag_us"] is None or doc["loop_lag_us"]["n"] >= 1
+
+
+def test_cpu_affinity_parse_and_apply(tmp_path):
+    """cpu_affinity spec parsing ("0", "0-1", "0,2") + application via
+    sched_setaffinity; invalid specs are logged, never fatal."""
+    before = os.sched_getaffinity(0)
+    try:
+        PluginManager._pin_cpus("0")
+        assert os.sched_getaffinity(0) == {0}
+        all_spec = ",".join(str(c) for c in sorted(before))
+        PluginManager._pin_cpus(all_spec)
+        assert os.sched_getaffinity(0) == before
+        PluginManager._pin_cpus("not-a-cpu")   # must not raise
+        PluginManager._pin_cpus("")            # no-op
+        assert os.sched_getaffinity(0) == before
+    finally:
+        os.sched_setaffinity(0, before)
