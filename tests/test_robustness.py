@@ -630,3 +630,39 @@ def test_watcher_mode_kubelet_flow(tmp_path):
     finally:
         ino.close()
         mgr.stop()
+
+
+def test_rescan_dynamic_plugin_watcher_mode(tmp_path):
+    """Dynamic rescan in registration_mode=watcher: the new *_VF plugin
+    exposes its pluginregistration socket under plugins_registry/ (the
+    modern-kubelet discovery path needs no kubelet.sock registration)."""
+    import tempfile as _tf
+    node = make_mock_node(str(tmp_path), n_gpus=1, kfd=False, hint=True)
+    registry = _tf.mkdtemp(prefix="kxdp-reg-")
+    cfg = node.config(registration_mode="watcher",
+                      plugins_registry_dir=registry)
+    mgr = PluginManager(cfg)
+    mgr.setup()
+    mgr.start(register=True)
+    try:
+        socks = set(os.listdir(registry))
+        assert len(socks) == 1
+        for k in range(2):
+            node.add_gpu(MockGPU(bdf=f"0000:61:02.{k}", device_id=0x75B3,
+                                 iommu_group=str(130 + k),
+                                 physfn_bdf="0000:61:00.0"))
+        assert mgr.rescan() is True
+        new_socks = set(os.listdir(registry)) - socks
+        assert len(new_socks) == 1
+        assert "_vf" in next(iter(new_socks))
+        # the registry socket serves the pluginregistration service
+        import grpc as _grpc
+        from kata_xpu_device_plugin_amd.plugin.watcher_registration import (
+            InfoRequest, WatcherRegistrationStub)
+        ch = _grpc.insecure_channel(
+            f"unix://{os.path.join(registry, next(iter(new_socks)))}")
+        info = WatcherRegistrationStub(ch).GetInfo(InfoRequest(), timeout=5)
+        assert info.name == "amd.com/INSTINCT_MI355X_VF"
+        ch.close()
+    finally:
+        mgr.stop()
