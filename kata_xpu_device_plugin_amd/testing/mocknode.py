@@ -43,6 +43,11 @@ class MockGPU:
     physfn_bdf: Optional[str] = None   # set ⇒ VF
     sriov_totalvfs: int = 0
     class_code: int = GPU_CLASS
+    # amdgpu partitioning sysfs (MI300/MI355 PFs; "" = files absent)
+    compute_partition: str = ""        # e.g. "SPX"
+    compute_available: str = ""        # e.g. "SPX, DPX, QPX, CPX"
+    memory_partition: str = ""         # e.g. "NPS1"
+    memory_available: str = ""         # e.g. "NPS1, NPS4"
 
 
 def default_bdfs(n: int) -> List[str]:
@@ -110,6 +115,16 @@ class MockNode:
         if gpu.sriov_totalvfs:
             self._write(os.path.join(d, "sriov_totalvfs"), f"{gpu.sriov_totalvfs}\n")
             self._write(os.path.join(d, "sriov_numvfs"), "0\n")
+        if gpu.compute_partition:
+            self._write(os.path.join(d, "current_compute_partition"),
+                        f"{gpu.compute_partition}\n")
+            self._write(os.path.join(d, "available_compute_partition"),
+                        f"{gpu.compute_available or gpu.compute_partition}\n")
+        if gpu.memory_partition:
+            self._write(os.path.join(d, "current_memory_partition"),
+                        f"{gpu.memory_partition}\n")
+            self._write(os.path.join(d, "available_memory_partition"),
+                        f"{gpu.memory_available or gpu.memory_partition}\n")
         if gpu.driver:
             self._symlink(
                 os.path.join(self.sysfs, "bus", "pci", "drivers", gpu.driver),

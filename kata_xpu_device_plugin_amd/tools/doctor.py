@@ -49,11 +49,18 @@ def diagnose(cfg: Config, probe: bool = False) -> dict:
     if not gpus:
         doc["problems"].append("no AMD GPU-class PCI functions on this node")
         return doc
-    doc["gpus"] = [
-        {"bdf": f.bdf, "device": f"{f.device:04x}", "driver": f.driver,
-         "iommu_group": f.iommu_group, "numa": f.numa_node, "vf": f.is_vf}
-        for f in gpus
-    ]
+    from .partition import read_partition_state
+    doc["gpus"] = []
+    for f in gpus:
+        entry = {"bdf": f.bdf, "device": f"{f.device:04x}", "driver": f.driver,
+                 "iommu_group": f.iommu_group, "numa": f.numa_node,
+                 "vf": f.is_vf}
+        if not f.is_vf:
+            pst = read_partition_state(cfg, f.bdf, f.driver)
+            if pst.supported:
+                entry["partition"] = {"compute": pst.compute_current,
+                                      "memory": pst.memory_current}
+        doc["gpus"].append(entry)
 
     topo = load_topology(cfg, inv)
     doc["topology"] = {

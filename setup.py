@@ -71,6 +71,7 @@ if __name__ == "__main__" or "setuptools" in sys.modules:
                     "kxdp-doctor=kata_xpu_device_plugin_amd.tools.doctor:main",
                     "kxdp-assignments=kata_xpu_device_plugin_amd.tools.assignments:main",
                     "kxdp-resourceslice=kata_xpu_device_plugin_amd.tools.resourceslice:main",
+                    "kxdp-partition=kata_xpu_device_plugin_amd.tools.partition:main",
                 ],
             },
             ext_modules=[

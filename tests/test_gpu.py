@@ -164,6 +164,30 @@ def test_amdsmi_live_xgmi_telemetry():
             assert not dh.xgmi_sick
 
 
+def test_partition_state_live_node():
+    """Partition sysfs on live MI355X (amdgpu-bound): current compute mode
+    is one of the defined modes and is in the advertised available set
+    (read-only — never mutate a shared lease box)."""
+    from kata_xpu_device_plugin_amd.config import Config
+    from kata_xpu_device_plugin_amd.tools.partition import (
+        COMPUTE_MODES, MEMORY_MODES, list_states)
+
+    states = [s for s in list_states(Config()) if s.driver == "amdgpu"]
+    assert states, "no amdgpu-bound PFs visible"
+    supported = [s for s in states if s.supported]
+    for st in supported:
+        if st.compute_current:
+            assert st.compute_current in COMPUTE_MODES, st
+            if st.compute_available:
+                assert st.compute_current in st.compute_available
+        if st.memory_current:
+            assert st.memory_current in MEMORY_MODES, st
+    # informational: record what the silicon exposes
+    print("partition states:", [
+        (s.bdf, s.compute_current, s.compute_available,
+         s.memory_current, s.memory_available) for s in states])
+
+
 def test_doctor_live_node():
     """kxdp-doctor on the live box: amdgpu-bound GPUs → pre-provisioning
     verdict (exit 0) with real discovery/topology/amd-smi content."""

@@ -108,7 +108,7 @@ def test_all_cli_entrypoints_have_help():
     any tool would break provisioning scripts)."""
     import subprocess
     import sys
-    mods = ["topo", "burnin", "ident", "bind", "sriov", "validate",
+    mods = ["topo", "burnin", "ident", "bind", "sriov", "validate", "partition",
             "assignments", "resourceslice", "doctor"]
     for m in mods:
         out = subprocess.run(
@@ -187,3 +187,57 @@ def test_daemonset_manifest_sane():
     vols = {v["name"] for v in spec["volumes"]}
     assert {m["name"] for m in ctr["volumeMounts"]} <= vols
     assert ctr["livenessProbe"]["httpGet"]["path"] == "/metrics"
+
+
+# --- partition tool (MI355X compute/memory partitioning) -------------------
+
+def _part_node(tmp_path):
+    from kata_xpu_device_plugin_amd.testing.mocknode import MockNode, MockGPU
+    node = MockNode(root=str(tmp_path))
+    node.add_gpu(MockGPU(bdf="0000:0a:00.0", driver="amdgpu", iommu_group="70",
+                         compute_partition="SPX",
+                         compute_available="SPX, DPX, QPX, CPX",
+                         memory_partition="NPS1",
+                         memory_available="NPS1, NPS4"))
+    node.add_gpu(MockGPU(bdf="0000:12:00.0", iommu_group="71"))  # vfio, no files
+    return node
+
+
+def test_partition_show(tmp_path, capsys):
+    import json as _json
+    from kata_xpu_device_plugin_amd.tools.partition import list_states, main
+    node = _part_node(tmp_path)
+    cfg = node.config()
+    states = {s.bdf: s for s in list_states(cfg)}
+    assert states["0000:0a:00.0"].compute_current == "SPX"
+    assert states["0000:0a:00.0"].compute_available == ["SPX", "DPX", "QPX", "CPX"]
+    assert states["0000:0a:00.0"].memory_current == "NPS1"
+    assert not states["0000:12:00.0"].supported   # vfio-bound: no files
+    rc = main(["--sysfs-root", cfg.sysfs_root, "--dev-root", cfg.dev_root,
+               "show"])
+    assert rc == 0
+    doc = _json.loads(capsys.readouterr().out)
+    assert {d["bdf"] for d in doc} == {"0000:0a:00.0", "0000:12:00.0"}
+
+
+def test_partition_set_paths(tmp_path):
+    import pytest as _pytest
+    from kata_xpu_device_plugin_amd.tools.partition import (
+        read_partition_state, set_partition)
+    node = _part_node(tmp_path)
+    cfg = node.config()
+    # happy path: amdgpu-bound, mode in available set
+    set_partition(cfg, "0000:0a:00.0", compute="CPX")
+    assert read_partition_state(cfg, "0000:0a:00.0").compute_current == "CPX"
+    set_partition(cfg, "0000:0a:00.0", memory="NPS4")
+    assert read_partition_state(cfg, "0000:0a:00.0").memory_current == "NPS4"
+    # refusals: vfio-bound GPU, unsupported mode, unknown bdf
+    with _pytest.raises(ValueError, match="amdgpu"):
+        set_partition(cfg, "0000:12:00.0", compute="CPX")
+    with _pytest.raises(ValueError, match="available"):
+        set_partition(cfg, "0000:0a:00.0", compute="TPX")
+    with _pytest.raises(ValueError, match="not an AMD"):
+        set_partition(cfg, "0000:ff:00.0", compute="CPX")
+    # dry run leaves state untouched
+    set_partition(cfg, "0000:0a:00.0", compute="SPX", dry_run=True)
+    assert read_partition_state(cfg, "0000:0a:00.0").compute_current == "CPX"
