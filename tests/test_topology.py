@@ -183,3 +183,26 @@ def test_vf_inherits_pf_hive(tmp_path):
     pick = preferred_allocation(topo, inv, ids, [], 2)
     hives = {topo.hive(inv.devices[d].primary.bdf) for d in pick}
     assert len(hives) == 1
+
+
+def test_degraded_must_include_still_honored():
+    """kubelet may force (must_include) a degraded-fabric GPU; the
+    selector honors it and packs the remainder for locality."""
+    from kata_xpu_device_plugin_amd.topology.hive import (
+        GPUTopology, preferred_sets, score_set)
+    topo = GPUTopology(source="hint")
+    bdf_of = {}
+    for i in range(6):
+        bdf = f"0000:{10+i:02x}:00.0"
+        bdf_of[str(i)] = bdf
+        topo.hive_of[bdf] = "hive-1" if i < 3 else "hive-2"
+    topo.set_degraded([bdf_of["0"]])
+    pick = preferred_sets(topo, bdf_of, list(bdf_of), ["0"], 3)
+    assert "0" in pick and len(pick) == 3
+    others = [d for d in pick if d != "0"]
+    hives = {topo.hive(bdf_of[d]) for d in others}
+    assert len(hives) == 1 and hives != {""}, \
+        "remaining picks must share one intact hive"
+    # degradation is reversible
+    topo.set_degraded([])
+    assert topo.hive(bdf_of["0"]) == "hive-1"
