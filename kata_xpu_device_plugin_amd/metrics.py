@@ -60,8 +60,10 @@ class MetricsExporter:
         m = self.manager
         if m.inventory is not None:
             self.g_scan_wall.set(m.inventory.scan_wall_s)
-        for rname, plugin in m.plugins.items():
-            state = m.states[rname]
+        for rname, plugin in list(m.plugins.items()):
+            state = m.states.get(rname)
+            if state is None:
+                continue
             self.g_devices.labels(rname).set(len(state.device_ids()))
             self.g_healthy.labels(rname).set(len(state.healthy_ids()))
             self.g_last_alloc.labels(rname).set(plugin.last_allocate_s)
@@ -86,7 +88,7 @@ class MetricsExporter:
             "kxdp_allocate_seconds",
             "Cumulative server-side Allocate() handler time",
             labels=["resource"])
-        for rname, plugin in self.manager.plugins.items():
+        for rname, plugin in list(self.manager.plugins.items()):
             c_allocs.add_metric([rname], plugin.allocations)
             c_fail.add_metric([rname], plugin.allocate_failures)
             c_secs.add_metric([rname], plugin.allocate_seconds_total)

@@ -647,3 +647,42 @@ def test_cpu_affinity_parse_and_apply(tmp_path):
         assert os.sched_getaffinity(0) == before
     finally:
         os.sched_setaffinity(0, before)
+
+
+def test_metrics_scrape_concurrent_with_rescan(tmp_path):
+    """/metrics scrapes race rescan's plugin-dict mutation (dynamic
+    start/retire): no 'dict changed size' crashes, scrape always returns."""
+    from prometheus_client import generate_latest
+    from kata_xpu_device_plugin_amd.metrics import MetricsExporter
+
+    node = make_mock_node(str(tmp_path), n_gpus=2, kfd=False, hint=True)
+    mgr = PluginManager(node.config())
+    mgr.setup()
+    exp = MetricsExporter(mgr)
+    stop = threading.Event()
+    errs = []
+
+    def scraper():
+        try:
+            while not stop.is_set():
+                assert b"kxdp_devices" in generate_latest(exp.registry)
+        except Exception as e:  # pragma: no cover
+            errs.append(e)
+
+    t = threading.Thread(target=scraper)
+    t.start()
+    try:
+        for cycle in range(6):
+            for k in range(2):
+                node.add_gpu(MockGPU(bdf=f"0000:61:02.{k}", device_id=0x75B3,
+                                     iommu_group=str(140 + k),
+                                     physfn_bdf="0000:61:00.0"))
+            mgr.rescan()
+            for k in range(2):
+                node.remove_gpu(f"0000:61:02.{k}")
+            mgr.rescan()
+    finally:
+        stop.set()
+        t.join(timeout=5)
+        mgr.stop()
+    assert not errs
