@@ -392,7 +392,10 @@ std::vector<std::string> select_preferred(
 
     long nodes_visited = 0;
     std::function<void(int, int, int64_t)> dfs = [&](int i, int left, int64_t acc) {
-        if (++nodes_visited > MAX_SEARCH_NODES) return;  // keep best so far
+        // Budget only bites once a complete solution exists: the first
+        // greedy descent (<= nb nodes) always finishes, so a feasible
+        // request never degrades to {} (mirrors hive.py).
+        if (++nodes_visited > MAX_SEARCH_NODES && !best_take.empty()) return;
         if (left == 0) {
             int64_t pack = packing();
             if (best_take.empty() || acc > best_score ||
