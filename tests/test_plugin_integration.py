@@ -249,3 +249,23 @@ def test_concurrent_allocate_churn(running):
         t.join()
     assert not errs
     assert _first_plugin(mgr).allocations == 400
+
+
+def test_listandwatch_unsubscribes_on_cancel(running):
+    """Cancelled/closed streams must release their DeviceState callback
+    subscriptions (the soak's RSS bound depends on this)."""
+    node, cfg, stub, mgr = running
+    r = stub.wait_for_registration(1)[0]
+    ps = stub.plugin_stub(r.endpoint)
+    state = next(iter(mgr.states.values()))
+    base = len(state._callbacks)
+    streams = [ps.ListAndWatch(api.Empty()) for _ in range(5)]
+    for s in streams:
+        next(s)
+    assert len(state._callbacks) == base + 5
+    for s in streams:
+        s.cancel()
+    deadline = time.time() + 5
+    while len(state._callbacks) > base and time.time() < deadline:
+        time.sleep(0.05)
+    assert len(state._callbacks) == base, "stream subscriptions leaked"
