@@ -214,7 +214,9 @@ def preferred_sets(
     # buckets) lands on the concentration-optimal solution immediately,
     # which makes the branch-and-bound prune everything else. Best-fit
     # preference is handled by the packing term, not the search order.
-    keys = sorted(buckets, key=lambda k: (-len(buckets[k]), k))
+    # Secondary sort by NUMA so indistinguishable buckets sit adjacent
+    # (symmetry reduction below).
+    keys = sorted(buckets, key=lambda k: (-len(buckets[k]), k[1], k[0]))
     caps = [len(buckets[k]) for k in keys]
 
     # Forced members contribute fixed pair terms with each candidate bucket;
@@ -232,6 +234,31 @@ def preferred_sets(
         return aff
 
     affinities = [bucket_affinity(k) for k in keys]
+
+    # ---- symmetry reduction -------------------------------------------
+    # Two whole-hive buckets are INDISTINGUISHABLE to the objective when
+    # they have equal capacity, the same NUMA node, the same must-set
+    # affinity, and each hive lives entirely in its one bucket: any
+    # assignment permuting their takes scores identically (pairs, packing
+    # and cross terms all match). Searching only the canonical
+    # non-increasing take order turns compositions over k identical hives
+    # into partitions — e.g. 8×8-VF hives at size 32 collapse from ~10^4
+    # equal-score compositions (tens of ms) to a handful.
+    hive_span: Dict[str, int] = {}
+    for k, cap in zip(keys, caps):
+        if k[0]:
+            hive_span[k[0]] = hive_span.get(k[0], 0) + cap
+    equiv = [False] * len(keys)
+    for i in range(1, len(keys)):
+        a, b = keys[i - 1], keys[i]
+        equiv[i] = (
+            bool(a[0]) and bool(b[0])
+            and caps[i] == caps[i - 1]
+            and a[1] == b[1]
+            and affinities[i] == affinities[i - 1]
+            and hive_span[a[0]] == caps[i - 1]
+            and hive_span[b[0]] == caps[i]
+        )
 
     # Packing tie-break: among equal-locality choices prefer the set that
     # leaves the least fragmentation — deplete small/partial hives first so
@@ -283,6 +310,12 @@ def preferred_sets(
             hive_cap_suffix[i][h] = hive_cap_suffix[i].get(h, 0) + caps[i]
         if n != -1:
             numa_cap_suffix[i][n] = numa_cap_suffix[i].get(n, 0) + caps[i]
+    # cap-descending group capacities per suffix, for the exact
+    # zero-taken concentration bound below
+    hive_caps_desc = [sorted(hive_cap_suffix[i].values(), reverse=True)
+                      for i in range(n_b + 1)]
+    numa_caps_desc = [sorted(numa_cap_suffix[i].values(), reverse=True)
+                      for i in range(n_b + 1)]
 
     best_score = -1
     best_packing = 0
@@ -292,29 +325,45 @@ def preferred_sets(
     taken_numa: Dict[int, int] = {}
     nodes_visited = 0
 
-    def xgmi_upper(i: int, left: int) -> int:
-        """Admissible bound on future xGMI pairs: the marginal gain of the
-        j-th future device placed into hive h (already holding t_h) is
-        t_h + j pairs; summing the `left` LARGEST marginals over all
-        remaining hives over-estimates any feasible placement (within a
-        hive marginals increase, so a top-k pick relaxes only the prefix
-        constraint — never under-counts)."""
+    def _concentration_exact(caps_desc: List[int], left: int) -> int:
+        """EXACT max of same-group pairs when every group starts empty:
+        fill largest-capacity groups fully, one partial remainder.
+        (Exchange argument: with c_a ≥ c_b, moving a device from b to a
+        gains c_a − (c_b − 1) > 0 pairs, so an optimum concentrates into
+        the largest groups.) Exact ⇒ admissible AND tight — this is what
+        lets symmetric many-hive requests prune at the root."""
+        total = 0
+        for cap in caps_desc:
+            if left <= 0:
+                break
+            c = cap if cap < left else left
+            total += c * (c - 1) // 2
+            left -= c
+        return total
+
+    def _topk_relaxed(cap_map: Dict, taken: Dict, left: int) -> int:
+        """Admissible bound with partially-taken groups: the j-th future
+        device into group g (already holding t_g) gains t_g + j pairs;
+        summing the `left` largest marginals over-estimates any feasible
+        placement (relaxes only the per-group prefix constraint)."""
         gains: List[int] = []
-        for h, cap in hive_cap_suffix[i].items():
-            t = taken_hive.get(h, 0)
+        for g, cap in cap_map.items():
+            t = taken.get(g, 0)
             gains.extend(range(t, t + cap))
         gains.sort(reverse=True)
         return sum(gains[:left])
 
+    def xgmi_upper(i: int, left: int) -> int:
+        cap_map = hive_cap_suffix[i]
+        if all(taken_hive.get(h, 0) == 0 for h in cap_map):
+            return _concentration_exact(hive_caps_desc[i], left)
+        return _topk_relaxed(cap_map, taken_hive, left)
+
     def numa_upper(i: int, left: int) -> int:
-        """Same marginal-gain bound for the NUMA tier (future-future and
-        future-placed same-NUMA pairs)."""
-        gains: List[int] = []
-        for n, cap in numa_cap_suffix[i].items():
-            t = taken_numa.get(n, 0)
-            gains.extend(range(t, t + cap))
-        gains.sort(reverse=True)
-        return sum(gains[:left])
+        cap_map = numa_cap_suffix[i]
+        if all(taken_numa.get(n, 0) == 0 for n in cap_map):
+            return _concentration_exact(numa_caps_desc[i], left)
+        return _topk_relaxed(cap_map, taken_numa, left)
 
     def dfs(i: int, left: int, acc: int):
         nonlocal best_score, best_packing, best_take, nodes_visited
@@ -339,7 +388,10 @@ def preferred_sets(
             if ub < best_score or (ub == best_score and best_packing == 0):
                 return
         hive, numa = keys[i]
-        for c in range(min(caps[i], left), -1, -1):
+        cmax = min(caps[i], left)
+        if equiv[i]:
+            cmax = min(cmax, take[i - 1])   # canonical non-increasing order
+        for c in range(cmax, -1, -1):
             inc = bucket_pair_score(keys[i], c) + affinities[i] * c
             if c:
                 if hive:

@@ -291,10 +291,12 @@ std::vector<std::string> select_preferred(
     // buckets) is the concentration optimum, so the branch-and-bound
     // prunes nearly everything else; best-fit preference comes from the
     // packing term, not search order (mirror of hive.py).
+    // secondary sort by NUMA then hive so indistinguishable buckets sit
+    // adjacent (symmetry reduction below; mirror of hive.py)
     std::sort(buckets.begin(), buckets.end(), [](const Bucket& a, const Bucket& b) {
         if (a.ids.size() != b.ids.size()) return a.ids.size() > b.ids.size();
-        if (a.hive != b.hive) return a.hive < b.hive;
-        return a.numa < b.numa;
+        if (a.numa != b.numa) return a.numa < b.numa;
+        return a.hive < b.hive;
     });
 
     const int nb = (int)buckets.size();
@@ -314,6 +316,24 @@ std::vector<std::string> select_preferred(
     std::vector<int> caps(nb), suffix(nb + 1, 0);
     for (int i = 0; i < nb; i++) caps[i] = (int)buckets[i].ids.size();
     for (int i = nb - 1; i >= 0; i--) suffix[i] = suffix[i + 1] + caps[i];
+
+    // ---- symmetry reduction (mirror of hive.py): whole-hive buckets
+    // with equal capacity, same NUMA, same must-affinity, whose hives
+    // live entirely in their one bucket, are interchangeable — search
+    // only canonical non-increasing takes across equivalent runs.
+    std::map<std::string, int> hive_span;
+    for (int i = 0; i < nb; i++)
+        if (!buckets[i].hive.empty()) hive_span[buckets[i].hive] += caps[i];
+    std::vector<char> equiv(nb, 0);
+    for (int i = 1; i < nb; i++) {
+        const Bucket& a = buckets[i - 1];
+        const Bucket& b = buckets[i];
+        equiv[i] = !a.hive.empty() && !b.hive.empty() &&
+                   caps[i] == caps[i - 1] && a.numa == b.numa &&
+                   affinity[i] == affinity[i - 1] &&
+                   hive_span[a.hive] == caps[i - 1] &&
+                   hive_span[b.hive] == caps[i];
+    }
 
     // packing-hive key per bucket: hives pack together; hiveless pack by numa
     auto pkey = [&](int i) -> std::pair<std::string, long> {
@@ -370,7 +390,35 @@ std::vector<std::string> select_preferred(
         for (int g : gains) sum += g;
         return sum;
     };
+    // EXACT tier max when every remaining group starts empty: fill
+    // largest-capacity groups fully (exchange argument) — tight, so the
+    // first greedy descent prunes symmetric requests at the root.
+    auto concentration_exact = [](const std::vector<int>& caps_desc,
+                                  int left) -> int64_t {
+        int64_t total = 0;
+        for (int cap : caps_desc) {
+            if (left <= 0) break;
+            int64_t c = cap < left ? cap : left;
+            total += c * (c - 1) / 2;
+            left -= (int)c;
+        }
+        return total;
+    };
+    // per-suffix cap-descending group capacity lists
+    std::vector<std::vector<int>> hive_caps_desc(nb + 1), numa_caps_desc(nb + 1);
+    for (int i = 0; i <= nb; i++) {
+        for (auto& kv : hive_cap_suffix[i]) hive_caps_desc[i].push_back(kv.second);
+        for (auto& kv : numa_cap_suffix[i]) numa_caps_desc[i].push_back(kv.second);
+        std::sort(hive_caps_desc[i].rbegin(), hive_caps_desc[i].rend());
+        std::sort(numa_caps_desc[i].rbegin(), numa_caps_desc[i].rend());
+    }
     auto xgmi_upper = [&](int i, int left) -> int64_t {
+        bool untouched = true;
+        for (auto& kv : hive_cap_suffix[i]) {
+            auto it = taken_hive.find(kv.first);
+            if (it != taken_hive.end() && it->second > 0) { untouched = false; break; }
+        }
+        if (untouched) return concentration_exact(hive_caps_desc[i], left);
         std::vector<int> gains;
         for (auto& kv : hive_cap_suffix[i]) {
             auto it = taken_hive.find(kv.first);
@@ -381,6 +429,12 @@ std::vector<std::string> select_preferred(
     };
     // same marginal-gain bound for the NUMA tier
     auto numa_upper = [&](int i, int left) -> int64_t {
+        bool untouched = true;
+        for (auto& kv : numa_cap_suffix[i]) {
+            auto it = taken_numa.find(kv.first);
+            if (it != taken_numa.end() && it->second > 0) { untouched = false; break; }
+        }
+        if (untouched) return concentration_exact(numa_caps_desc[i], left);
         std::vector<int> gains;
         for (auto& kv : numa_cap_suffix[i]) {
             auto it = taken_numa.find(kv.first);
@@ -415,7 +469,9 @@ std::vector<std::string> select_preferred(
         }
         const std::string& hive = buckets[i].hive;
         const long numa = buckets[i].numa;
-        for (int c = std::min(caps[i], left); c >= 0; c--) {
+        int cmax = std::min(caps[i], left);
+        if (equiv[i]) cmax = std::min(cmax, take[i - 1]);  // canonical order
+        for (int c = cmax; c >= 0; c--) {
             int64_t inc = bucket_pair(i, c) + affinity[i] * c;
             if (c) {
                 if (!hive.empty()) inc += W_XGMI * c * taken_hive[hive];
