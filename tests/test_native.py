@@ -147,7 +147,11 @@ def test_selector_vf_scale_bounded():
         pick = preferred_sets(topo, bdf_of, ids, [], k, use_native=True)
         elapsed = time.perf_counter() - t0
         assert len(pick) == k
-        assert elapsed < 0.2, f"selection at k={k} took {elapsed:.3f}s"
+        # round 2: symmetry reduction + exact concentration bound put the
+        # worst case (~73 us on MI355X hosts) far under this ceiling even
+        # on slow shared CI (was 23 ms before — 0.2 s ceiling kept 10x
+        # slack for CI noise, now 100x)
+        assert elapsed < 0.05, f"selection at k={k} took {elapsed:.3f}s"
         # xGMI tier must be concentration-optimal: ceil(k/8) hives
         hives = {topo.hive(bdf_of[d]) for d in pick}
         assert len(hives) == (k + 7) // 8
