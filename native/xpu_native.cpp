@@ -376,19 +376,40 @@ std::vector<std::string> select_preferred(
         for (auto& kv : taken) p -= hive_free[kv.first] - kv.second;
         return p;
     };
-    // Admissible bound on future xGMI pairs: marginal gain of the j-th
-    // future device in hive h (holding t) is t+j; the sum of the `left`
-    // largest marginals over all remaining hives over-estimates every
-    // feasible placement (mirror of hive.py::xgmi_upper).
-    auto top_marginals = [](std::vector<int>& gains, int left) -> int64_t {
-        if ((int)gains.size() > left) {
-            std::partial_sort(gains.begin(), gains.begin() + left, gains.end(),
-                              std::greater<int>());
-            gains.resize(left);
+    // Admissible bound on future pairs: marginal gain of the j-th future
+    // device in group g (holding t) is t+j; the sum of the `left` largest
+    // marginals over-estimates every feasible placement. Each group's
+    // marginals are the integer range [t, t+cap), so the top-k sum is a
+    // THRESHOLD binary search over ranges — O(H log V) per node instead
+    // of materializing every marginal (mirror of hive.py::_topk_relaxed).
+    auto topk_ranges = [](const std::vector<std::pair<int, int>>& ranges,
+                          int left) -> int64_t {
+        if (ranges.empty() || left <= 0) return 0;
+        int lo = 0, hi = 0;
+        for (auto& r : ranges) hi = std::max(hi, r.first + r.second);
+        while (lo < hi) {
+            int mid = (lo + hi + 1) / 2;
+            long cnt = 0;
+            for (auto& r : ranges) {
+                int top = r.first + r.second;
+                if (top > mid) cnt += top - std::max(mid, r.first);
+            }
+            if (cnt >= left) lo = mid; else hi = mid - 1;
         }
-        int64_t sum = 0;
-        for (int g : gains) sum += g;
-        return sum;
+        const int T = lo;
+        int64_t total = 0;
+        long cnt_above = 0;
+        for (auto& r : ranges) {
+            int top = r.first + r.second;
+            int start = std::max(T + 1, r.first);
+            if (top > start) {
+                int64_t k = top - start;
+                total += k * (start + top - 1) / 2;
+                cnt_above += k;
+            }
+        }
+        total += (int64_t)(left - cnt_above) * T;
+        return total;
     };
     // EXACT tier max when every remaining group starts empty: fill
     // largest-capacity groups fully (exchange argument) — tight, so the
@@ -419,13 +440,13 @@ std::vector<std::string> select_preferred(
             if (it != taken_hive.end() && it->second > 0) { untouched = false; break; }
         }
         if (untouched) return concentration_exact(hive_caps_desc[i], left);
-        std::vector<int> gains;
+        std::vector<std::pair<int, int>> ranges;
         for (auto& kv : hive_cap_suffix[i]) {
             auto it = taken_hive.find(kv.first);
             int t = it != taken_hive.end() ? it->second : 0;
-            for (int j = 0; j < kv.second; j++) gains.push_back(t + j);
+            ranges.emplace_back(t, kv.second);
         }
-        return top_marginals(gains, left);
+        return topk_ranges(ranges, left);
     };
     // same marginal-gain bound for the NUMA tier
     auto numa_upper = [&](int i, int left) -> int64_t {
@@ -435,13 +456,13 @@ std::vector<std::string> select_preferred(
             if (it != taken_numa.end() && it->second > 0) { untouched = false; break; }
         }
         if (untouched) return concentration_exact(numa_caps_desc[i], left);
-        std::vector<int> gains;
+        std::vector<std::pair<int, int>> ranges;
         for (auto& kv : numa_cap_suffix[i]) {
             auto it = taken_numa.find(kv.first);
             int t = it != taken_numa.end() ? it->second : 0;
-            for (int j = 0; j < kv.second; j++) gains.push_back(t + j);
+            ranges.emplace_back(t, kv.second);
         }
-        return top_marginals(gains, left);
+        return topk_ranges(ranges, left);
     };
 
     long nodes_visited = 0;
