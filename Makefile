@@ -1,7 +1,7 @@
 PYTHON ?= python3
 IMAGE  ?= ghcr.io/example/kata-xpu-device-plugin-amd:0.1.0
 
-.PHONY: build build-hip test test-gpu bench image clean doctor validate burnin
+.PHONY: build build-hip test test-gpu bench soak image clean doctor validate burnin partition
 
 build:
 	$(PYTHON) setup.py build_ext --inplace
@@ -26,6 +26,12 @@ validate:
 
 burnin:
 	$(PYTHON) -m kata_xpu_device_plugin_amd.tools.burnin
+
+partition:
+	$(PYTHON) -m kata_xpu_device_plugin_amd.tools.partition show
+
+soak:
+	$(PYTHON) benchmarks/churn_soak.py --minutes 10 --clients 3
 
 image:
 	docker build -t $(IMAGE) .
