@@ -62,6 +62,11 @@ def _build_fdp() -> descriptor_pb2.FileDescriptorProto:
 
     m = msg("ListPodResourcesResponse")
     fld(m, "pod_resources", 1, _F.TYPE_MESSAGE, _F.LABEL_REPEATED, "PodResources")
+
+    msg("AllocatableResourcesRequest")
+    m = msg("AllocatableResourcesResponse")
+    fld(m, "devices", 1, _F.TYPE_MESSAGE, _F.LABEL_REPEATED, "ContainerDevices")
+    fld(m, "cpu_ids", 2, _F.TYPE_INT64, _F.LABEL_REPEATED)
     return fdp
 
 
@@ -74,6 +79,8 @@ def _cls(name):
 
 
 ListPodResourcesRequest = _cls("ListPodResourcesRequest")
+AllocatableResourcesRequest = _cls("AllocatableResourcesRequest")
+AllocatableResourcesResponse = _cls("AllocatableResourcesResponse")
 ContainerDevices = _cls("ContainerDevices")
 ContainerResources = _cls("ContainerResources")
 PodResources = _cls("PodResources")
@@ -83,6 +90,7 @@ _SERVICE = f"{_PKG}.PodResourcesLister"
 # kubelet serves "v1.PodResourcesLister"; our package name must match the
 # wire path, so the real client overrides the service path:
 KUBELET_SERVICE_PATH = "/v1.PodResourcesLister/List"
+KUBELET_ALLOCATABLE_PATH = "/v1.PodResourcesLister/GetAllocatableResources"
 
 
 @dataclass
@@ -122,6 +130,26 @@ class PodResourcesClient:
                     ))
         return out
 
+    def allocatable(self, resource_prefix: str = "amd.com/") -> Dict[str, List[str]]:
+        """resource name → device ids kubelet considers allocatable
+        (GetAllocatableResources — capacity view, complements list())."""
+        ch = grpc.insecure_channel(f"unix://{self.socket_path}")
+        try:
+            grpc.channel_ready_future(ch).result(timeout=self.timeout_s)
+            call = ch.unary_unary(
+                KUBELET_ALLOCATABLE_PATH,
+                request_serializer=AllocatableResourcesRequest.SerializeToString,
+                response_deserializer=AllocatableResourcesResponse.FromString,
+            )
+            resp = call(AllocatableResourcesRequest(), timeout=self.timeout_s)
+        finally:
+            ch.close()
+        out: Dict[str, List[str]] = {}
+        for dev in resp.devices:
+            if dev.resource_name.startswith(resource_prefix):
+                out.setdefault(dev.resource_name, []).extend(dev.device_ids)
+        return out
+
     def assignments(self, resource_prefix: str = "amd.com/") -> Dict[str, str]:
         """device id → 'namespace/pod/container' for our resources."""
         out: Dict[str, str] = {}
@@ -141,6 +169,12 @@ def add_lister_servicer(server: grpc.Server, servicer, service_name: str = "v1.P
             response_serializer=ListPodResourcesResponse.SerializeToString,
         ),
     }
+    if hasattr(servicer, "GetAllocatableResources"):
+        handlers["GetAllocatableResources"] = grpc.unary_unary_rpc_method_handler(
+            servicer.GetAllocatableResources,
+            request_deserializer=AllocatableResourcesRequest.FromString,
+            response_serializer=AllocatableResourcesResponse.SerializeToString,
+        )
     server.add_generic_rpc_handlers(
         (grpc.method_handlers_generic_handler(service_name, handlers),)
     )

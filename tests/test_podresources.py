@@ -84,3 +84,38 @@ def test_assignments_filtering(lister_socket):
     client = pr.PodResourcesClient(lister_socket)
     m = client.assignments("amd.com/")
     assert m == {"70": "ml/train-0/worker", "71": "ml/train-0/worker"}
+
+
+class _ListerWithAllocatable(_Lister):
+    def GetAllocatableResources(self, request, context):
+        return pr.AllocatableResourcesResponse(devices=[
+            pr.ContainerDevices(resource_name="amd.com/INSTINCT_MI355X",
+                                device_ids=["70", "71", "72"]),
+            pr.ContainerDevices(resource_name="other.io/dev",
+                                device_ids=["z"]),
+        ])
+
+
+def test_allocatable_roundtrip():
+    """GetAllocatableResources over a real unix socket: the capacity view
+    complementing List (kubelet podresources v1)."""
+    sock_dir = tempfile.mkdtemp(prefix="kxdp-podres-")
+    path = os.path.join(sock_dir, "kubelet.sock")
+    server = grpc.server(futures.ThreadPoolExecutor(max_workers=2))
+    pr.add_lister_servicer(server, _ListerWithAllocatable())
+    server.add_insecure_port(f"unix://{path}")
+    server.start()
+    try:
+        client = pr.PodResourcesClient(socket_path=path, timeout_s=5)
+        alloc = client.allocatable()
+        assert alloc == {"amd.com/INSTINCT_MI355X": ["70", "71", "72"]}
+    finally:
+        server.stop(grace=None)
+
+
+def test_allocatable_wire_format():
+    """AllocatableResourcesResponse field numbers pinned (devices=1)."""
+    resp = pr.AllocatableResourcesResponse(devices=[
+        pr.ContainerDevices(resource_name="amd.com/X", device_ids=["1"])])
+    dev = _ld(1, b"amd.com/X") + _ld(2, b"1")
+    assert resp.SerializeToString() == _ld(1, dev)
