@@ -19,10 +19,13 @@ def main(argv=None) -> int:
     p = argparse.ArgumentParser(prog="kxdp-assignments")
     p.add_argument("--socket", default=DEFAULT_SOCKET)
     p.add_argument("--prefix", default="amd.com/")
+    p.add_argument("--allocatable", action="store_true",
+                   help="show kubelet's allocatable capacity view instead "
+                        "of current pod assignments")
     args = p.parse_args(argv)
     client = PodResourcesClient(args.socket)
     try:
-        data = client.assignments(args.prefix)
+        data = client.allocatable(args.prefix) if args.allocatable             else client.assignments(args.prefix)
     except Exception as e:
         print(f"cannot query pod-resources at {args.socket}: {e}", file=sys.stderr)
         return 1
