@@ -25,7 +25,8 @@ def main(argv=None) -> int:
     args = p.parse_args(argv)
     client = PodResourcesClient(args.socket)
     try:
-        data = client.allocatable(args.prefix) if args.allocatable             else client.assignments(args.prefix)
+        data = (client.allocatable(args.prefix) if args.allocatable
+                else client.assignments(args.prefix))
     except Exception as e:
         print(f"cannot query pod-resources at {args.socket}: {e}", file=sys.stderr)
         return 1
